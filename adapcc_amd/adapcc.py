@@ -1,0 +1,99 @@
+"""AdapCC facade (reference: adapcc.py:15-76).
+
+Class-level API kept call-compatible with the reference so its usage recipes
+port 1:1:
+
+    AdapCC.init(args, local_rank, world_rank, world_size)
+    AdapCC.setup(Primitive.ALLREDUCE)
+    AdapCC.communicator.all_reduce(tensor)
+    AdapCC.reconstruct_topology()
+    AdapCC.clear()
+
+plus the DDP hook in adapcc_amd.runtime.hook.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Sequence
+
+import torch
+
+from .communicator import CommArgs, Communicator
+from .primitives import Primitive
+
+
+class AdapCC:
+    communicator: Optional[Communicator] = None
+
+    @classmethod
+    def init(cls, args, local_rank: int, world_rank: int, world_size: int,
+             group=None) -> None:
+        if not isinstance(args, CommArgs):
+            args = CommArgs.from_namespace(args)
+        cls.communicator = Communicator(args, local_rank, world_rank,
+                                        world_size, group=group)
+        cls.communicator.run_entry_point()
+
+    @classmethod
+    def setup(cls, primitive: Primitive = Primitive.ALLREDUCE) -> None:
+        cls._require_init()
+        cls.communicator.setup(primitive)
+
+    @classmethod
+    def allreduce(cls, tensor: torch.Tensor,
+                  active: Optional[Sequence[int]] = None,
+                  average: bool = False) -> torch.Tensor:
+        cls._require_init()
+        return cls.communicator.all_reduce(tensor, active=active, average=average)
+
+    # reference spelling aliases
+    all_reduce = allreduce
+
+    @classmethod
+    def reduce(cls, tensor: torch.Tensor, root: int = 0,
+               active: Optional[Sequence[int]] = None) -> torch.Tensor:
+        cls._require_init()
+        return cls.communicator.reduce(tensor, root=root, active=active)
+
+    @classmethod
+    def boardcast(cls, tensor: torch.Tensor, root: int = 0) -> torch.Tensor:
+        cls._require_init()
+        return cls.communicator.broadcast(tensor, root=root)
+
+    broadcast = boardcast
+
+    @classmethod
+    def allgather(cls, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        cls._require_init()
+        return cls.communicator.all_gather(out, tensor)
+
+    @classmethod
+    def alltoall(cls, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        cls._require_init()
+        return cls.communicator.all_to_all(out, tensor)
+
+    @classmethod
+    def reducescatter(cls, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        cls._require_init()
+        return cls.communicator.reduce_scatter(out, tensor)
+
+    @classmethod
+    def update_relay(cls, step: int) -> None:
+        cls._require_init()
+        cls.communicator.update_relay(step)
+
+    @classmethod
+    def reconstruct_topology(cls) -> None:
+        cls._require_init()
+        cls.communicator.reconstruct_topology()
+
+    @classmethod
+    def clear(cls) -> None:
+        if cls.communicator is not None:
+            cls.communicator.clear()
+            cls.communicator = None
+
+    @classmethod
+    def _require_init(cls) -> None:
+        if cls.communicator is None:
+            raise RuntimeError("AdapCC.init() has not been called")

@@ -1,0 +1,271 @@
+"""Control plane (reference: commu.py CudaCommu).
+
+Owns the engine (native xGMI pull-engine on GPU, process-group fallback on
+CPU or by explicit opt-in), the adaptation flow (detect -> profile ->
+synthesize), relay control, and the DDP hook state.
+
+Differences from the reference by design:
+ - bootstrap/artifact exchange goes over torch.distributed object
+   collectives instead of scp + ctypes + sleep-based readiness
+   (ref commu.py:318 time_init_wait)
+ - one engine serves all primitives (no per-primitive contexts/ports)
+ - transport selection: ADAPCC_TRANSPORT = native | pg | auto (default auto:
+   native on GPU, with a guarded self-test at setup; pg on CPU)
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional, Sequence
+
+import torch
+import torch.distributed as dist
+
+from .primitives import Primitive
+from .strategy.synthesizer import Synthesizer
+from .topology.formats import (
+    LogicalGraph,
+    ProfileMatrices,
+    Strategy,
+    load_strategy,
+    single_node_graph,
+)
+
+log = logging.getLogger("adapcc")
+
+
+@dataclass
+class CommArgs:
+    """The launcher flag contract (reference: launcher.py:54-62)."""
+
+    port: int = 18000
+    entry_point: int = -1
+    strategy_file: str = ""
+    logical_graph: str = ""
+    parallel_degree: int = 0  # 0 = auto (world_size stars on one node)
+    profile_freq: int = 0     # reconstruct_topology every N steps (0 = never)
+    policy: str = "par-trees"
+    chunk_bytes: int = 4 * 1024 * 1024
+
+    @classmethod
+    def from_namespace(cls, ns) -> "CommArgs":
+        kw = {}
+        for f in ("port", "entry_point", "strategy_file", "logical_graph",
+                  "parallel_degree", "profile_freq", "policy", "chunk_bytes"):
+            if hasattr(ns, f) and getattr(ns, f) is not None:
+                kw[f] = getattr(ns, f)
+        return cls(**kw)
+
+
+class Communicator:
+    def __init__(
+        self,
+        args: CommArgs,
+        local_rank: int,
+        world_rank: int,
+        world_size: int,
+        group=None,
+    ) -> None:
+        self.args = args
+        self.local_rank = local_rank
+        self.rank = world_rank
+        self.world_size = world_size
+        self.group = group
+        self.engine = None
+        self.strategy: Optional[Strategy] = None
+        self.graph: Optional[LogicalGraph] = None
+        self.profile_mats: Optional[ProfileMatrices] = None
+        self.transport = os.environ.get("ADAPCC_TRANSPORT", "auto")
+        self.use_gpu = torch.cuda.is_available()
+        self.coordinator = None   # rank-0 gRPC server (relay/fault)
+        self.controller = None    # per-rank controller thread
+        self.active_ranks: Optional[List[int]] = None  # None = all
+        self._setup_done = False
+
+        self.synthesizer = Synthesizer(
+            policy=args.policy,
+            parallel_degree=args.parallel_degree or max(2, world_size),
+            chunk_bytes=args.chunk_bytes,
+        )
+
+    # ------------------------------------------------------------------
+    # Adaptation flow (reference: adapcc.py:16-41 init sequence)
+    # ------------------------------------------------------------------
+
+    def run_entry_point(self) -> None:
+        ep = self.args.entry_point
+        if ep == int(Primitive.DETECT):
+            self.detect_topology()
+            self.profile_topology()
+            self.synthesize()
+        elif ep == int(Primitive.PROFILE):
+            self.load_or_default_graph()
+            self.profile_topology()
+            self.synthesize()
+        else:
+            if self.args.strategy_file and os.path.exists(self.args.strategy_file):
+                self.strategy = load_strategy(self.args.strategy_file)
+            else:
+                self.load_or_default_graph()
+                self.synthesize()
+
+    def load_or_default_graph(self) -> None:
+        if self.args.logical_graph and os.path.exists(self.args.logical_graph):
+            from .topology.formats import load_logical_graph
+
+            self.graph = load_logical_graph(self.args.logical_graph)
+        else:
+            self.graph = single_node_graph(self.world_size)
+
+    def detect_topology(self) -> None:
+        from .topology.detect import detect_node_topology
+
+        self.graph = detect_node_topology(
+            self.rank, self.local_rank, self.world_size, group=self.group
+        )
+
+    def profile_topology(self) -> None:
+        from .topology.profile import profile_links
+
+        t0 = time.time()
+        self.profile_mats = profile_links(
+            self.rank, self.world_size, self.graph, group=self.group
+        )
+        log.info("profiling time: %.3f s", time.time() - t0)
+
+    def synthesize(self) -> None:
+        if self.graph is None:
+            self.load_or_default_graph()
+        self.strategy = self.synthesizer.generate_strategy(
+            graph=self.graph, profile=self.profile_mats
+        )
+
+    # ------------------------------------------------------------------
+    # Engine lifecycle (reference: commu.py:301-352 init/exit_threads)
+    # ------------------------------------------------------------------
+
+    def setup(self, primitive: Primitive = Primitive.ALLREDUCE) -> None:
+        if self.strategy is None:
+            self.run_entry_point()
+        t0 = time.time()
+        transport = self.transport
+        if transport == "auto":
+            transport = "native" if self.use_gpu else "pg"
+
+        if transport == "native":
+            from .runtime.engine import NativeEngine
+
+            torch.cuda.set_device(self.local_rank)
+            self.engine = NativeEngine(self.rank, self.world_size,
+                                       device=self.local_rank)
+            self.engine.bootstrap(group=self.group)
+            self.engine.set_strategy(self.strategy)
+            self.engine.self_test()
+        elif transport == "pg":
+            from .runtime.fallback import ProcessGroupEngine
+
+            self.engine = ProcessGroupEngine(self.rank, self.world_size,
+                                             group=self.group)
+            self.engine.set_strategy(self.strategy)
+        else:
+            raise ValueError(f"unknown ADAPCC_TRANSPORT={transport}")
+        self._setup_done = True
+        log.info("[Rank %d] transmission context setup time: %.1f ms",
+                 self.rank, 1000 * (time.time() - t0))
+
+    def ensure_setup(self) -> None:
+        if not self._setup_done:
+            self.setup()
+
+    # ------------------------------------------------------------------
+    # Collectives (reference: commu.py:360-379)
+    # ------------------------------------------------------------------
+
+    def all_reduce(self, tensor: torch.Tensor,
+                   active: Optional[Sequence[int]] = None,
+                   average: bool = False) -> torch.Tensor:
+        self.ensure_setup()
+        if active is None:
+            active = self.active_ranks
+        return self.engine.all_reduce(tensor, active=active, average=average)
+
+    def synchronize(self) -> None:
+        if self.engine is not None:
+            self.engine.synchronize()
+
+    # The remaining primitives currently route through the native engine when
+    # it implements them, else torch.distributed (RCCL/gloo). The tree engine
+    # gains native reduce/broadcast/allgather/alltoall incrementally.
+
+    def reduce(self, tensor: torch.Tensor, root: int = 0,
+               active: Optional[Sequence[int]] = None) -> torch.Tensor:
+        self.ensure_setup()
+        if hasattr(self.engine, "reduce"):
+            return self.engine.reduce(tensor, root=root, active=active)
+        if active is not None and len(active) > 0 and self.rank not in set(active):
+            tensor.zero_()
+        dist.reduce(tensor, dst=root, group=self.group)
+        return tensor
+
+    def broadcast(self, tensor: torch.Tensor, root: int = 0) -> torch.Tensor:
+        self.ensure_setup()
+        if hasattr(self.engine, "broadcast"):
+            return self.engine.broadcast(tensor, root=root)
+        dist.broadcast(tensor, src=root, group=self.group)
+        return tensor
+
+    def all_gather(self, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        self.ensure_setup()
+        if hasattr(self.engine, "all_gather"):
+            return self.engine.all_gather(out, tensor)
+        dist.all_gather_into_tensor(out, tensor, group=self.group)
+        return out
+
+    def reduce_scatter(self, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        self.ensure_setup()
+        if hasattr(self.engine, "reduce_scatter"):
+            return self.engine.reduce_scatter(out, tensor)
+        dist.reduce_scatter_tensor(out, tensor, group=self.group)
+        return out
+
+    def all_to_all(self, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        self.ensure_setup()
+        if hasattr(self.engine, "all_to_all"):
+            return self.engine.all_to_all(out, tensor)
+        dist.all_to_all_single(out, tensor, group=self.group)
+        return out
+
+    # ------------------------------------------------------------------
+    # On-the-fly re-adaptation (reference: adapcc.py:63-67)
+    # ------------------------------------------------------------------
+
+    def reconstruct_topology(self) -> None:
+        self.clear(keep_coordinator=True)
+        self.profile_topology()
+        self.synthesize()
+        self.setup()
+
+    def update_relay(self, step: int) -> None:
+        """Feed the controller a new step (relay negotiation happens in the
+        background controller thread; see coordinator/)."""
+        if self.controller is not None:
+            self.controller.submit_step(step)
+
+    def clear(self, keep_coordinator: bool = False) -> None:
+        if self.engine is not None:
+            try:
+                self.engine.synchronize()
+            except Exception:
+                pass
+            self.engine = None
+        self._setup_done = False
+        if not keep_coordinator:
+            if self.controller is not None:
+                self.controller.stop()
+                self.controller = None
+            if self.coordinator is not None:
+                self.coordinator.stop()
+                self.coordinator = None

@@ -1,0 +1,127 @@
+// pybind11 bindings for the adapcc_amd native engine.
+//
+// Deliberately torch-free at the C++ level: tensors arrive as
+// (data_ptr, numel, dtype) and the HIP stream as an integer
+// (torch.cuda.current_stream().cuda_stream), so the extension builds with
+// plain hipcc and the Python wrapper (adapcc_amd/runtime/engine.py) owns all
+// torch-level safety checks.
+
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "common.h"
+#include "engine.h"
+#include "plan.h"
+
+namespace py = pybind11;
+using adapcc::Engine;
+
+// Expose the unit-plan computation (plan.cpp) for CPU-side unit tests: the
+// Python test harness simulates the full multi-rank protocol over these
+// plans with numpy buffers, covering relay control and deadlock-freedom
+// without a GPU.
+static py::dict plan_to_dict(const adapcc::PlanData& p) {
+  py::list cs, rs, bs;
+  for (const auto& u : p.cunits) {
+    py::dict d;
+    d["tree"] = u.tree; d["chunk"] = u.chunk;
+    d["offset"] = u.offset_elems; d["count"] = u.count_elems;
+    d["notify"] = (bool)u.notify_parent; d["consumer"] = u.parent_rank;
+    cs.append(d);
+  }
+  for (const auto& u : p.runits) {
+    py::dict d;
+    d["tree"] = u.tree; d["chunk"] = u.chunk;
+    d["offset"] = u.offset_elems; d["count"] = u.count_elems;
+    py::list srcs;
+    for (int s = 0; s < u.nsrc; ++s)
+      srcs.append(py::make_tuple(u.src_rank[s], (int)u.src_kind[s]));
+    d["srcs"] = srcs;
+    d["include_self"] = (bool)u.include_self;
+    d["notify"] = (bool)u.notify_parent; d["consumer"] = u.parent_rank;
+    d["is_root"] = (bool)u.is_root;
+    py::list kids;
+    for (int c = 0; c < u.nchildren; ++c) kids.append(u.child_rank[c]);
+    d["publish_to"] = kids;
+    rs.append(d);
+  }
+  for (const auto& u : p.bunits) {
+    py::dict d;
+    d["tree"] = u.tree; d["chunk"] = u.chunk;
+    d["offset"] = u.offset_elems; d["count"] = u.count_elems;
+    d["parent"] = u.parent_rank; d["parent_kind"] = (int)u.parent_kind;
+    d["forward"] = (bool)u.forward;
+    py::list kids;
+    for (int c = 0; c < u.nchildren; ++c) kids.append(u.child_rank[c]);
+    d["publish_to"] = kids;
+    bs.append(d);
+  }
+  py::dict out;
+  out["copy"] = cs;
+  out["reduce"] = rs;
+  out["bcast"] = bs;
+  out["chunk_elems"] = p.chunk_elems;
+  return out;
+}
+
+PYBIND11_MODULE(_core, m) {
+  m.doc() = "adapcc_amd native engine (MI355X / gfx950)";
+
+  m.def("compute_plan",
+        [](const std::vector<std::vector<int>>& parents, int rank,
+           long total_elems, int esize, long chunk_bytes,
+           const std::vector<int>& active) {
+          auto shape = adapcc::TreeShape::derive(parents);
+          uint64_t mask = 0;
+          for (int r : active) mask |= (1ull << r);
+          if (active.empty()) mask = (1ull << shape.world) - 1;
+          return plan_to_dict(adapcc::build_plan(shape, rank, total_elems,
+                                                 esize, chunk_bytes, mask));
+        },
+        py::arg("parents"), py::arg("rank"), py::arg("total_elems"),
+        py::arg("esize"), py::arg("chunk_bytes"),
+        py::arg("active") = std::vector<int>{});
+
+  py::class_<Engine>(m, "Engine")
+      .def(py::init<int, int, int, size_t, double>(), py::arg("rank"),
+           py::arg("world"), py::arg("device"), py::arg("cap_bytes"),
+           py::arg("timeout_ms"))
+      .def("ipc_handle",
+           [](Engine& e) { return py::bytes(e.ipc_handle()); })
+      .def("connect",
+           [](Engine& e, const std::vector<py::bytes>& handles) {
+             std::vector<std::string> hs;
+             hs.reserve(handles.size());
+             for (const auto& h : handles) hs.push_back(std::string(h));
+             e.connect(hs);
+           })
+      .def("set_strategy", &Engine::set_strategy, py::arg("parents"),
+           py::arg("chunk_bytes"))
+      .def("allreduce",
+           [](Engine& e, uintptr_t data_ptr, long numel, int dtype, int op,
+              const std::vector<int>& active, bool average,
+              uintptr_t stream) {
+             e.allreduce(reinterpret_cast<void*>(data_ptr), numel, dtype, op,
+                         active, average, reinterpret_cast<void*>(stream));
+           },
+           py::arg("data_ptr"), py::arg("numel"), py::arg("dtype"),
+           py::arg("op"), py::arg("active"), py::arg("average"),
+           py::arg("stream"))
+      .def("synchronize", &Engine::synchronize,
+           py::call_guard<py::gil_scoped_release>())
+      .def("query_error", &Engine::query_error)
+      .def_property_readonly("rank", &Engine::rank)
+      .def_property_readonly("world", &Engine::world)
+      .def_property_readonly("capacity", &Engine::capacity)
+      .def_property_readonly("num_trees", &Engine::num_trees);
+
+  m.attr("DTYPE_F32") = (int)adapcc::Dtype::F32;
+  m.attr("DTYPE_F16") = (int)adapcc::Dtype::F16;
+  m.attr("DTYPE_BF16") = (int)adapcc::Dtype::BF16;
+  m.attr("OP_SUM") = (int)adapcc::RedOp::Sum;
+  m.attr("OP_AVG") = (int)adapcc::RedOp::Avg;
+  m.attr("OP_MAX") = (int)adapcc::RedOp::Max;
+  m.attr("OP_MIN") = (int)adapcc::RedOp::Min;
+  m.attr("MAX_RANKS") = adapcc::kMaxRanks;
+  m.attr("MAX_TREES") = adapcc::kMaxTrees;
+}

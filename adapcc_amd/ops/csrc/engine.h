@@ -1,0 +1,102 @@
+// Engine class declaration (see engine.hip for design notes).
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <map>
+#include <string>
+#include <tuple>
+#include <utility>
+#include <vector>
+
+#include "common.h"
+#include "plan.h"
+
+namespace adapcc {
+
+class Engine {
+ public:
+  Engine(int rank, int world, int device, size_t cap_bytes, double timeout_ms);
+  ~Engine();
+  Engine(const Engine&) = delete;
+  Engine& operator=(const Engine&) = delete;
+
+  std::string ipc_handle() const;
+  void connect(const std::vector<std::string>& handles);
+  void set_strategy(const std::vector<std::vector<int>>& parents,
+                    long chunk_bytes);
+
+  // Enqueue an allreduce of `total_elems` elements of `dtype` at `data` on
+  // the caller stream. active_ranks empty => all ranks. average => divide
+  // by |active| on the broadcast write.
+  void allreduce(void* data, long total_elems, int dtype, int op,
+                 const std::vector<int>& active_ranks, bool average,
+                 void* caller_stream);
+
+  void synchronize();
+  std::pair<uint64_t, uint64_t> query_error();
+
+  int rank() const { return rank_; }
+  int world() const { return world_; }
+  size_t capacity() const { return cap_bytes_; }
+  int num_trees() const { return num_trees_; }
+
+  struct Plan {
+    std::vector<CopyUnit> cunits;
+    std::vector<ReduceUnit> runits;
+    std::vector<BcastUnit> bunits;
+    CopyUnit* d_c = nullptr;
+    ReduceUnit* d_r = nullptr;
+    BcastUnit* d_b = nullptr;
+    int* d_ranks = nullptr;
+    int nranks = 0;
+    long total_elems = 0;
+    Dtype dt = Dtype::F32;
+    void free_device();
+  };
+
+ private:
+  struct PlanKey {
+    long elems;
+    int dt;
+    int op;
+    uint64_t mask;
+    bool operator<(const PlanKey& o) const {
+      return std::tie(elems, dt, op, mask) < std::tie(o.elems, o.dt, o.op, o.mask);
+    }
+  };
+
+  void build_tables();
+  Plan& get_plan(long total_elems, Dtype dt, RedOp op, uint64_t active_mask);
+
+  int rank_, world_, device_;
+  size_t cap_bytes_;
+  double timeout_ms_;
+  uint64_t seq_ = 0;
+  bool connected_ = false;
+
+  void* region_ = nullptr;
+  size_t region_bytes_ = 0;
+  size_t inbox_off_ = 0;
+  void* peer_base_[kMaxRanks];
+
+  DevTables tabs_{};
+
+  unsigned long long* counters_ = nullptr;
+  uint64_t* h_err_ = nullptr;
+
+  hipStream_t s_red_{}, s_bcast_{}, s_err_{};
+  hipEvent_t ev_in_{}, ev_sync0_{}, ev_red_{}, ev_barrier_{};
+
+  int num_trees_ = 0;
+  long chunk_bytes_ = 4 * 1024 * 1024;
+  TreeShape shape_;
+
+  int wgs_per_group_ = 8;
+  int n_groups_ = 16;
+
+  std::map<PlanKey, Plan> plans_;
+};
+
+}  // namespace adapcc

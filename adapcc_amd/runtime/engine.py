@@ -1,0 +1,172 @@
+"""Python wrapper around the native engine (`adapcc_amd._core`).
+
+Owns all torch-level safety: dtype mapping, contiguity, capacity splitting,
+bootstrap handle exchange over torch.distributed, and the loud failure when
+the native extension is missing on a GPU machine (never a silent eager
+fallback).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import List, Optional, Sequence
+
+import torch
+
+from ..topology.formats import Strategy
+
+_DTYPE_MAP = {}
+
+
+def _core():
+    try:
+        from adapcc_amd import _core as core  # noqa
+    except ImportError as e:  # pragma: no cover
+        raise RuntimeError(
+            "adapcc_amd._core native extension not built; run "
+            "`python -m adapcc_amd.ops.build` (hipcc, gfx950)"
+        ) from e
+    return core
+
+
+def _dtype_code(dtype: torch.dtype) -> int:
+    core = _core()
+    if not _DTYPE_MAP:
+        _DTYPE_MAP.update({
+            torch.float32: core.DTYPE_F32,
+            torch.float16: core.DTYPE_F16,
+            torch.bfloat16: core.DTYPE_BF16,
+        })
+    if dtype not in _DTYPE_MAP:
+        raise TypeError(f"unsupported dtype for adapcc engine: {dtype}")
+    return _DTYPE_MAP[dtype]
+
+
+def strategy_parent_arrays(strategy: Strategy, world_size: int) -> List[List[int]]:
+    """Convert the XML-level forest into per-tree parent arrays."""
+    out = []
+    for tree in strategy.trees:
+        parents = [-2] * world_size
+        stack = [(tree, -1)]
+        while stack:
+            node, par = stack.pop()
+            if node.rank >= world_size or node.rank < 0:
+                raise ValueError(f"rank {node.rank} out of range")
+            if parents[node.rank] != -2:
+                raise ValueError(f"rank {node.rank} appears twice in a tree")
+            parents[node.rank] = par
+            for c in node.children:
+                stack.append((c, node.rank))
+        if any(p == -2 for p in parents):
+            missing = [r for r, p in enumerate(parents) if p == -2]
+            raise ValueError(f"tree does not cover ranks {missing}")
+        out.append(parents)
+    return out
+
+
+class NativeEngine:
+    """The hipIpc/xGMI pull-engine. One instance per process (= per GPU)."""
+
+    def __init__(
+        self,
+        rank: int,
+        world_size: int,
+        device: Optional[int] = None,
+        cap_bytes: Optional[int] = None,
+        timeout_ms: Optional[float] = None,
+    ) -> None:
+        core = _core()
+        if device is None:
+            device = torch.cuda.current_device()
+        if cap_bytes is None:
+            cap_bytes = int(os.environ.get("ADAPCC_BUF_CAP", 512 * 1024 * 1024))
+        if timeout_ms is None:
+            timeout_ms = float(os.environ.get("ADAPCC_TIMEOUT_MS", 30000.0))
+        self.rank = rank
+        self.world_size = world_size
+        self.device = device
+        self.cap_bytes = cap_bytes
+        self._eng = core.Engine(rank, world_size, device, cap_bytes, timeout_ms)
+        self._connected = world_size == 1
+        self._strategy_set = False
+
+    # -- bootstrap ---------------------------------------------------------
+
+    def bootstrap(self, group=None) -> None:
+        """Exchange hipIpc handles through torch.distributed (any backend)."""
+        if self.world_size == 1:
+            self._connected = True
+            return
+        import torch.distributed as dist
+
+        handle = self._eng.ipc_handle()
+        gathered: List[Optional[bytes]] = [None] * self.world_size
+        dist.all_gather_object(gathered, handle, group=group)
+        self._eng.connect(list(gathered))
+        self._connected = True
+
+    def set_strategy(self, strategy: Strategy) -> None:
+        parents = strategy_parent_arrays(strategy, self.world_size)
+        self._eng.set_strategy(parents, strategy.chunk_bytes)
+        self._strategy_set = True
+
+    # -- collectives -------------------------------------------------------
+
+    def all_reduce(
+        self,
+        tensor: torch.Tensor,
+        active: Optional[Sequence[int]] = None,
+        average: bool = False,
+    ) -> torch.Tensor:
+        """Enqueue an in-place allreduce on the current stream. Asynchronous:
+        the caller stream is made to depend on completion."""
+        if self.world_size == 1:
+            return tensor
+        if not self._connected:
+            raise RuntimeError("engine not bootstrapped")
+        if not self._strategy_set:
+            raise RuntimeError("no strategy set")
+        if not tensor.is_contiguous():
+            raise ValueError("adapcc all_reduce requires a contiguous tensor")
+        if not tensor.is_cuda:
+            raise ValueError("adapcc native engine requires a GPU tensor")
+        core = _core()
+        dt = _dtype_code(tensor.dtype)
+        op = core.OP_AVG if average else core.OP_SUM
+        esize = tensor.element_size()
+        active_list = list(active) if active is not None else []
+        stream = torch.cuda.current_stream(tensor.device).cuda_stream
+
+        max_elems = self.cap_bytes // esize
+        numel = tensor.numel()
+        if numel <= max_elems:
+            self._eng.allreduce(tensor.data_ptr(), numel, dt, op, active_list,
+                                average, stream)
+        else:
+            flat = tensor.view(-1)
+            for beg in range(0, numel, max_elems):
+                piece = flat[beg : beg + max_elems]
+                self._eng.allreduce(piece.data_ptr(), piece.numel(), dt, op,
+                                    active_list, average, stream)
+        return tensor
+
+    def synchronize(self) -> None:
+        self._eng.synchronize()
+
+    def query_error(self):
+        return self._eng.query_error()
+
+    def self_test(self) -> None:
+        """Tiny correctness check (reference: adapcc.py:106-115 golden run).
+        Raises on mismatch or timeout."""
+        if self.world_size == 1:
+            return
+        t = torch.full((4096,), float(self.rank + 1), device=f"cuda:{self.device}")
+        self.all_reduce(t)
+        self.synchronize()
+        expect = sum(range(1, self.world_size + 1))
+        if not torch.allclose(t, torch.full_like(t, float(expect))):
+            got = t[0].item()
+            raise RuntimeError(
+                f"engine self-test mismatch: expected {expect}, got {got}"
+            )

@@ -1,0 +1,69 @@
+"""DDP gradient-bucket communication hook (reference: commu.py:385-435).
+
+Replaces RCCL's bucket allreduce inside PyTorch DDP with the adapcc engine.
+The hook averages over the *active* set (relay control): stragglers excluded
+from the active set neither contribute nor aggregate, but still receive the
+reduced gradients (a stronger consistency model than the reference's BSP
+mode, which let inactive replicas silently diverge — commu.py:424-431).
+
+Usage:
+    state = AdapccDDPState(AdapCC.communicator)
+    model.register_comm_hook(state, adapcc_allreduce_hook)
+    ...
+    state.on_step(step)  # per-iteration relay update
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+import torch
+
+from ..communicator import Communicator
+
+
+@dataclass
+class AdapccDDPState:
+    comm: Communicator
+    step: int = 0
+    bucket_elems: List[int] = field(default_factory=list)
+    _first_bucket_of_step: bool = True
+
+    def on_step(self, step: int) -> None:
+        """Call once per iteration before backward (reference:
+        train_ddp.py:43 update_relay)."""
+        self.step = step
+        self._first_bucket_of_step = True
+        self.comm.update_relay(step)
+
+    @property
+    def active(self) -> Optional[List[int]]:
+        return self.comm.active_ranks
+
+
+def adapcc_allreduce_hook(state: AdapccDDPState, bucket) -> torch.futures.Future:
+    tensor = bucket.buffer()
+    if state._first_bucket_of_step:
+        state._first_bucket_of_step = False
+        if hasattr(state.comm, "notify_hook_ready"):
+            state.comm.notify_hook_ready(state.step)
+        if state.step == 1 and hasattr(bucket, "index"):
+            pass  # bucket layout is stable from step 1 on; nothing to record
+    n_active = (len(state.active) if state.active else state.comm.world_size)
+    state.comm.all_reduce(tensor, active=state.active, average=True)
+    fut: torch.futures.Future = torch.futures.Future()
+    fut.set_result(tensor)
+    return fut
+
+
+def adapcc_bf16_compress_hook(state: AdapccDDPState, bucket) -> torch.futures.Future:
+    """bf16-compressed variant: halves xGMI traffic for fp32 buckets.
+    (PyTorch parity: ddp_comm_hooks.default_hooks.bf16_compress_hook.)"""
+    buf = bucket.buffer()
+    half = buf.to(torch.bfloat16)
+    state.comm.all_reduce(half, active=state.active, average=True)
+    fut: torch.futures.Future = torch.futures.Future()
+    buf.copy_(half.to(buf.dtype))
+    fut.set_result(buf)
+    return fut

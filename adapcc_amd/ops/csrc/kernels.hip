@@ -406,6 +406,69 @@ __global__ void barrier_kernel(DevTables tabs, CallArgs args, int me, int world,
 }
 
 // ---------------------------------------------------------------------------
+// Standalone multi-source reduction (no flags/tables): dst = op over nsrc
+// local buffers. Used for single-GPU numerics tests of the reduction path
+// and by fused ops (e.g. MoE combine). Same vectorized inner loop as the
+// pull-reduce.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void __launch_bounds__(256) local_reduce_kernel(
+    T* __restrict__ dst, const T* const* __restrict__ srcs, int nsrc,
+    long count, RedOp op, float scale) {
+  using VT = VecTraits<T>;
+  using Vec = typename VT::Vec;
+  const long stride = (long)gridDim.x * blockDim.x;
+  const long tid = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long vcount = count / VT::kPerVec;
+  for (long i = tid; i < vcount; i += stride) {
+    float acc[VT::kPerVec];
+    Vec v = reinterpret_cast<const Vec*>(srcs[0])[i];
+    VT::unpack(v, acc);
+    for (int s = 1; s < nsrc; ++s) {
+      float tmp[VT::kPerVec];
+      Vec w = reinterpret_cast<const Vec*>(srcs[s])[i];
+      VT::unpack(w, tmp);
+#pragma unroll
+      for (int k = 0; k < VT::kPerVec; ++k) acc[k] = red_combine(op, acc[k], tmp[k]);
+    }
+#pragma unroll
+    for (int k = 0; k < VT::kPerVec; ++k) acc[k] *= scale;
+    reinterpret_cast<Vec*>(dst)[i] = VT::pack(acc);
+  }
+  for (long j = vcount * VT::kPerVec + tid; j < count; j += stride) {
+    float a = VT::load1(srcs[0] + j);
+    for (int s = 1; s < nsrc; ++s) a = red_combine(op, a, VT::load1(srcs[s] + j));
+    VT::store1(dst + j, a * scale);
+  }
+}
+
+void launch_local_reduce(Dtype dt, void* dst, const void* const* srcs_dev,
+                         int nsrc, long count, RedOp op, float scale,
+                         hipStream_t stream) {
+  const dim3 block(256);
+  const dim3 grid(512);
+  switch (dt) {
+    case Dtype::F32:
+      hipLaunchKernelGGL((local_reduce_kernel<float>), grid, block, 0, stream,
+                         (float*)dst, (const float* const*)srcs_dev, nsrc, count,
+                         op, scale);
+      break;
+    case Dtype::BF16:
+      hipLaunchKernelGGL((local_reduce_kernel<__hip_bfloat16>), grid, block, 0,
+                         stream, (__hip_bfloat16*)dst,
+                         (const __hip_bfloat16* const*)srcs_dev, nsrc, count, op,
+                         scale);
+      break;
+    case Dtype::F16:
+      hipLaunchKernelGGL((local_reduce_kernel<__half>), grid, block, 0, stream,
+                         (__half*)dst, (const __half* const*)srcs_dev, nsrc,
+                         count, op, scale);
+      break;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // Launchers
 // ---------------------------------------------------------------------------
 

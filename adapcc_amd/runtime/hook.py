@@ -13,8 +13,6 @@ Usage:
     state.on_step(step)  # per-iteration relay update
 """
 
-from __future__ import annotations
-
 from dataclasses import dataclass, field
 from typing import List, Optional
 
@@ -42,7 +40,9 @@ class AdapccDDPState:
         return self.comm.active_ranks
 
 
-def adapcc_allreduce_hook(state: AdapccDDPState, bucket) -> torch.futures.Future:
+def adapcc_allreduce_hook(
+    state: AdapccDDPState, bucket
+) -> torch.futures.Future[torch.Tensor]:
     tensor = bucket.buffer()
     if state._first_bucket_of_step:
         state._first_bucket_of_step = False
@@ -57,7 +57,9 @@ def adapcc_allreduce_hook(state: AdapccDDPState, bucket) -> torch.futures.Future
     return fut
 
 
-def adapcc_bf16_compress_hook(state: AdapccDDPState, bucket) -> torch.futures.Future:
+def adapcc_bf16_compress_hook(
+    state: AdapccDDPState, bucket
+) -> torch.futures.Future[torch.Tensor]:
     """bf16-compressed variant: halves xGMI traffic for fp32 buckets.
     (PyTorch parity: ddp_comm_hooks.default_hooks.bf16_compress_hook.)"""
     buf = bucket.buffer()

@@ -1,0 +1,167 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GPT-2 small DDP training step over the adapcc engine.
+
+Measures whole-job tokens/sec (bf16 autocast, synthetic tokens, random-init
+weights) with gradient buckets allreduced by the adapcc_amd engine via the
+DDP communication hook — the reference's train_ddp.py workload re-targeted
+at GPT-2 small per BASELINE.json.
+
+Launch (driver contract):
+    python bench.py --gpus 1 --steps K --warmup W
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+        --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+import torch.distributed as dist
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from adapcc_amd import AdapCC, CommArgs  # noqa: E402
+from adapcc_amd.models.gpt2 import GPT2, GPT2Config  # noqa: E402
+from adapcc_amd.runtime.hook import AdapccDDPState, adapcc_allreduce_hook  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=10)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--batch", type=int, default=16, help="per-GPU micro batch")
+    p.add_argument("--seq", type=int, default=1024)
+    p.add_argument("--model", type=str, default="small",
+                   choices=["tiny", "small", "medium"])
+    p.add_argument("--bucket-mb", type=int, default=100)
+    p.add_argument("--device", type=str, default=None)
+    p.add_argument("--transport", type=str, default=None,
+                   help="override ADAPCC_TRANSPORT")
+    return p.parse_args()
+
+
+def main() -> None:
+    args = parse_args()
+    if args.transport:
+        os.environ["ADAPCC_TRANSPORT"] = args.transport
+
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+
+    use_cuda = torch.cuda.is_available() and args.device != "cpu"
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    if use_cuda:
+        torch.cuda.set_device(device)
+
+    if world > 1:
+        dist.init_process_group(
+            backend="nccl" if use_cuda else "gloo", rank=rank, world_size=world
+        )
+
+    torch.manual_seed(1234 + rank)
+    cfg = getattr(GPT2Config, args.model)()
+    model = GPT2(cfg).to(device)
+    n_params = model.num_params()
+
+    AdapCC.init(CommArgs(entry_point=-1, policy="par-trees"),
+                local_rank, rank, world)
+    AdapCC.setup()
+
+    if world > 1:
+        ddp = torch.nn.parallel.DistributedDataParallel(
+            model, device_ids=[local_rank] if use_cuda else None,
+            bucket_cap_mb=args.bucket_mb,
+        )
+        state = AdapccDDPState(AdapCC.communicator)
+        ddp.register_comm_hook(state, adapcc_allreduce_hook)
+        train_mod = ddp
+    else:
+        state = None
+        train_mod = model
+
+    opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4, betas=(0.9, 0.95),
+                            weight_decay=0.1)
+
+    B, T = args.batch, min(args.seq, cfg.n_positions)
+    data = torch.randint(0, cfg.vocab_size, (B, T + 1), device=device)
+    x, y = data[:, :-1], data[:, 1:].contiguous()
+
+    amp_dtype = torch.bfloat16
+    autocast = torch.autocast(device_type=device.type, dtype=amp_dtype,
+                              enabled=True)
+
+    def step(i: int) -> None:
+        if state is not None:
+            state.on_step(i)
+        opt.zero_grad(set_to_none=True)
+        with autocast:
+            _, loss = train_mod(x, y)
+        loss.backward()
+        opt.step()
+
+    for i in range(args.warmup):
+        step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if world > 1:
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu",
+                         dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    tokens = world * B * T * args.steps
+    toks_per_s = tokens / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "gpt2_ddp_tokens_per_sec",
+            "value": toks_per_s,
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": f"gpt2-{args.model}-{n_params/1e6:.0f}M",
+                "global_batch": world * B,
+                "seq_len": T,
+                "parallelism": f"dp{world}",
+                "bucket_cap_mb": args.bucket_mb,
+                "transport": os.environ.get("ADAPCC_TRANSPORT", "auto"),
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    AdapCC.clear()
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,243 @@
+"""Multi-process CPU tests of the control plane over gloo (world_size 2-4):
+facade init/setup/allreduce, relay semantics, DDP hook integration, profile
+module. These exercise everything except the HIP data plane (covered by the
+plan simulator on CPU and by -m gpu tests on hardware)."""
+
+import os
+
+import pytest
+import torch
+
+from util_mp import run_mp
+
+
+def _facade_allreduce(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    from adapcc_amd import AdapCC, CommArgs
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+    t = torch.full((1000,), float(rank + 1))
+    AdapCC.allreduce(t)
+    AdapCC.communicator.synchronize()
+    expect = sum(range(1, world + 1))
+    assert torch.allclose(t, torch.full_like(t, float(expect)))
+    AdapCC.clear()
+    return True
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_facade_allreduce(world):
+    assert all(run_mp(_facade_allreduce, world))
+
+
+def _relay_active_subset(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    from adapcc_amd import AdapCC, CommArgs
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+    active = [r for r in range(world) if r != 1]
+    t = torch.full((64,), float(rank + 1))
+    AdapCC.allreduce(t, active=active, average=True)
+    expect = sum(r + 1 for r in active) / len(active)
+    assert torch.allclose(t, torch.full_like(t, expect)), t[0]
+    AdapCC.clear()
+    return True
+
+
+def test_relay_active_subset():
+    assert all(run_mp(_relay_active_subset, 3))
+
+
+def _ddp_hook_training(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    import torch.nn as nn
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.runtime.hook import AdapccDDPState, adapcc_allreduce_hook
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+
+    torch.manual_seed(7)  # same init on all ranks
+    model = nn.Sequential(nn.Linear(32, 64), nn.ReLU(), nn.Linear(64, 8))
+    ddp = DDP(model, bucket_cap_mb=1)
+    state = AdapccDDPState(AdapCC.communicator)
+    ddp.register_comm_hook(state, adapcc_allreduce_hook)
+    opt = torch.optim.SGD(ddp.parameters(), lr=0.1)
+
+    torch.manual_seed(100 + rank)  # different data per rank
+    losses = []
+    for step in range(3):
+        state.on_step(step)
+        x = torch.randn(16, 32)
+        y = torch.randn(16, 8)
+        opt.zero_grad()
+        loss = ((ddp(x) - y) ** 2).mean()
+        loss.backward()
+        opt.step()
+        losses.append(float(loss))
+    # replicas must stay in sync: compare a parameter hash across ranks
+    p = torch.cat([q.flatten() for q in model.parameters()])
+    import torch.distributed as dist
+
+    gathered = [torch.zeros_like(p) for _ in range(world)]
+    dist.all_gather(gathered, p)
+    for g in gathered:
+        assert torch.allclose(g, gathered[0], atol=1e-6)
+    AdapCC.clear()
+    return losses
+
+
+def test_ddp_hook_training():
+    res = run_mp(_ddp_hook_training, 2)
+    assert len(res[0]) == 3
+
+
+def _ddp_hook_matches_reference(rank, world):
+    """Gradients via adapcc hook == gradients via stock DDP allreduce."""
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    import torch.nn as nn
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.runtime.hook import AdapccDDPState, adapcc_allreduce_hook
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+
+    def make_model():
+        torch.manual_seed(7)
+        return nn.Sequential(nn.Linear(16, 16), nn.Tanh(), nn.Linear(16, 4))
+
+    def data():
+        torch.manual_seed(500 + rank)
+        return torch.randn(8, 16), torch.randn(8, 4)
+
+    # run A: adapcc hook
+    m1 = DDP(make_model(), bucket_cap_mb=1)
+    state = AdapccDDPState(AdapCC.communicator)
+    m1.register_comm_hook(state, adapcc_allreduce_hook)
+    x, y = data()
+    state.on_step(0)
+    ((m1(x) - y) ** 2).mean().backward()
+    g1 = torch.cat([p.grad.flatten() for p in m1.parameters()])
+
+    # run B: stock DDP
+    m2 = DDP(make_model(), bucket_cap_mb=1)
+    ((m2(x) - y) ** 2).mean().backward()
+    g2 = torch.cat([p.grad.flatten() for p in m2.parameters()])
+
+    assert torch.allclose(g1, g2, atol=1e-6), (g1 - g2).abs().max()
+    AdapCC.clear()
+    return True
+
+
+def test_ddp_hook_matches_reference():
+    assert all(run_mp(_ddp_hook_matches_reference, 2))
+
+
+def _profile_and_synthesize(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.primitives import Primitive
+
+    AdapCC.init(CommArgs(entry_point=int(Primitive.PROFILE)), rank, rank, world)
+    AdapCC.setup()
+    comm = AdapCC.communicator
+    assert comm.strategy is not None
+    comm.strategy.validate(world)
+    assert comm.profile_mats is not None
+    assert len(comm.profile_mats.bandwidth) == world * (world - 1)
+    t = torch.ones(128)
+    AdapCC.allreduce(t)
+    assert torch.allclose(t, torch.full_like(t, float(world)))
+    AdapCC.clear()
+    return True
+
+
+def test_profile_and_synthesize():
+    assert all(run_mp(_profile_and_synthesize, 2))
+
+
+def _detect_entry(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.primitives import Primitive
+
+    AdapCC.init(CommArgs(entry_point=int(Primitive.DETECT)), rank, rank, world)
+    AdapCC.setup()
+    comm = AdapCC.communicator
+    assert comm.graph is not None
+    assert comm.graph.ranks() == list(range(world))
+    AdapCC.clear()
+    return True
+
+
+def test_detect_entry():
+    assert all(run_mp(_detect_entry, 2))
+
+
+def _reconstruct_topology(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    from adapcc_amd import AdapCC, CommArgs
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+    t = torch.ones(64)
+    AdapCC.allreduce(t)
+    AdapCC.reconstruct_topology()
+    t2 = torch.ones(64)
+    AdapCC.allreduce(t2)
+    assert torch.allclose(t2, torch.full_like(t2, float(world)))
+    AdapCC.clear()
+    return True
+
+
+def test_reconstruct_topology():
+    assert all(run_mp(_reconstruct_topology, 2))
+
+
+def _other_primitives(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    from adapcc_amd import AdapCC, CommArgs
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+    comm = AdapCC.communicator
+
+    t = torch.full((32,), float(rank + 1))
+    comm.reduce(t, root=0)
+    if rank == 0:
+        assert torch.allclose(t, torch.full_like(t, float(sum(range(1, world + 1)))))
+
+    b = torch.full((32,), float(rank))
+    comm.broadcast(b, root=1)
+    assert torch.allclose(b, torch.full_like(b, 1.0))
+
+    out = torch.zeros(32 * world)
+    comm.all_gather(out, torch.full((32,), float(rank)))
+    for r in range(world):
+        assert torch.allclose(out[r * 32:(r + 1) * 32],
+                              torch.full((32,), float(r)))
+
+    src = torch.arange(world * 4, dtype=torch.float32) + rank * 1000
+    dst = torch.zeros(world * 4)
+    comm.all_to_all(dst, src)
+    for r in range(world):
+        expect = torch.arange(rank * 4, rank * 4 + 4, dtype=torch.float32) + r * 1000
+        assert torch.allclose(dst[r * 4:(r + 1) * 4], expect)
+
+    rs_in = torch.full((world * 8,), float(rank + 1))
+    rs_out = torch.zeros(8)
+    comm.reduce_scatter(rs_out, rs_in)
+    assert torch.allclose(rs_out, torch.full_like(rs_out, float(sum(range(1, world + 1)))))
+
+    AdapCC.clear()
+    return True
+
+
+def test_other_primitives():
+    assert all(run_mp(_other_primitives, 2))

@@ -128,6 +128,8 @@ PYBIND11_MODULE(_core, m) {
            })
       .def("set_strategy", &Engine::set_strategy, py::arg("parents"),
            py::arg("chunk_bytes"))
+      .def("connect_local", &Engine::connect_local, py::arg("peer_addrs"))
+      .def("region_addr", &Engine::region_addr)
       .def("allreduce",
            [](Engine& e, uintptr_t data_ptr, long numel, int dtype, int op,
               const std::vector<int>& active, bool average,

@@ -24,6 +24,10 @@ class Engine {
 
   std::string ipc_handle() const;
   void connect(const std::vector<std::string>& handles);
+  // Test/emulation path: wire peers by raw device address (same process,
+  // same device). Exercises the full flag protocol on one GPU.
+  void connect_local(const std::vector<uintptr_t>& peer_addrs);
+  uintptr_t region_addr() const { return (uintptr_t)region_; }
   void set_strategy(const std::vector<std::vector<int>>& parents,
                     long chunk_bytes);
 
@@ -75,6 +79,7 @@ class Engine {
   double timeout_ms_;
   uint64_t seq_ = 0;
   bool connected_ = false;
+  bool local_peers_ = false;
 
   void* region_ = nullptr;
   size_t region_bytes_ = 0;

@@ -21,6 +21,7 @@
 #include <cstring>
 #include <map>
 #include <mutex>
+#include <cstdlib>
 #include <stdexcept>
 #include <string>
 #include <tuple>
@@ -86,11 +87,17 @@ Engine::Engine(int rank, int world, int device, size_t cap_bytes,
   HIP_CHECK(hipHostMalloc(&h_err_, 2 * sizeof(uint64_t)));
   for (int r = 0; r < kMaxRanks; ++r) peer_base_[r] = nullptr;
   peer_base_[rank_] = region_;
+  if (const char* s = getenv("ADAPCC_WGS_PER_GROUP")) wgs_per_group_ = atoi(s);
+  if (const char* s = getenv("ADAPCC_N_GROUPS")) n_groups_ = atoi(s);
+  if (wgs_per_group_ < 1) wgs_per_group_ = 1;
+  if (n_groups_ < 1) n_groups_ = 1;
 }
 
 Engine::~Engine() {
-  for (int r = 0; r < world_; ++r) {
-    if (r != rank_ && peer_base_[r]) hipIpcCloseMemHandle(peer_base_[r]);
+  if (!local_peers_) {
+    for (int r = 0; r < world_; ++r) {
+      if (r != rank_ && peer_base_[r]) (void)hipIpcCloseMemHandle(peer_base_[r]);
+    }
   }
   if (region_) hipFree(region_);
   if (counters_) hipFree(counters_);
@@ -125,6 +132,18 @@ void Engine::connect(const std::vector<std::string>& handles) {
     HIP_CHECK(hipIpcOpenMemHandle(&p, h, hipIpcMemLazyEnablePeerAccess));
     peer_base_[r] = p;
   }
+  connected_ = true;
+  build_tables();
+}
+
+void Engine::connect_local(const std::vector<uintptr_t>& peer_addrs) {
+  if ((int)peer_addrs.size() != world_)
+    throw std::runtime_error("connect_local: need one address per rank");
+  for (int r = 0; r < world_; ++r) {
+    if (r == rank_) continue;
+    peer_base_[r] = reinterpret_cast<void*>(peer_addrs[r]);
+  }
+  local_peers_ = true;
   connected_ = true;
   build_tables();
 }

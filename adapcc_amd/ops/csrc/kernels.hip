@@ -191,8 +191,11 @@ __global__ void __launch_bounds__(256) copyin_kernel(
   for (int ui = group; ui < n_units; ui += n_groups) {
     const CopyUnit u = units[ui];
     const long cnt = u.count_elems;
-    // this wg's share, vector-aligned
-    const long per = (cnt + wgs_per_group - 1) / wgs_per_group;
+    // this wg's share, rounded up to the vector width so every non-final
+    // boundary stays 16-byte aligned
+    const long per =
+        ((cnt + wgs_per_group - 1) / wgs_per_group + VT::kPerVec - 1) /
+        VT::kPerVec * VT::kPerVec;
     const long beg = min((long)wg_in_group * per, cnt);
     const long end = min(beg + per, cnt);
     const T* src = user + u.offset_elems;
@@ -264,7 +267,9 @@ __global__ void __launch_bounds__(256) reduce_kernel(
 
     T* dst = (T*)tabs.acc[me] + u.offset_elems;
     const long cnt = u.count_elems;
-    const long per = (cnt + wgs_per_group - 1) / wgs_per_group;
+    const long per =
+        ((cnt + wgs_per_group - 1) / wgs_per_group + VT::kPerVec - 1) /
+        VT::kPerVec * VT::kPerVec;
     const long beg = min((long)wg_in_group * per, cnt);
     const long end = min(beg + per, cnt);
     const long vend = beg + ((end - beg) / VT::kPerVec) * VT::kPerVec;
@@ -339,7 +344,9 @@ __global__ void __launch_bounds__(256) bcast_kernel(
     T* fwd = u.forward ? (T*)tabs.result[me] + u.offset_elems : nullptr;
 
     const long cnt = u.count_elems;
-    const long per = (cnt + wgs_per_group - 1) / wgs_per_group;
+    const long per =
+        ((cnt + wgs_per_group - 1) / wgs_per_group + VT::kPerVec - 1) /
+        VT::kPerVec * VT::kPerVec;
     const long beg = min((long)wg_in_group * per, cnt);
     const long end = min(beg + per, cnt);
     const long vend = beg + ((end - beg) / VT::kPerVec) * VT::kPerVec;

@@ -143,6 +143,7 @@ PYBIND11_MODULE(_core, m) {
       .def("synchronize", &Engine::synchronize,
            py::call_guard<py::gil_scoped_release>())
       .def("query_error", &Engine::query_error)
+      .def("dump_inbox", [](Engine& e) { return py::bytes(e.dump_inbox()); })
       .def_property_readonly("rank", &Engine::rank)
       .def_property_readonly("world", &Engine::world)
       .def_property_readonly("capacity", &Engine::capacity)

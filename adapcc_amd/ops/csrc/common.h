@@ -44,6 +44,8 @@ struct FlagInbox {
   // kernel error/status word: 0 ok; else error code (host-polled)
   uint64_t error;
   uint64_t error_detail;
+  // debug trace slots (written by kernels; host-dumped)
+  uint64_t trace[32];
 };
 
 // Error codes written by kernels on bounded-spin timeout.

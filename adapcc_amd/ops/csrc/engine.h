@@ -40,6 +40,7 @@ class Engine {
 
   void synchronize();
   std::pair<uint64_t, uint64_t> query_error();
+  std::string dump_inbox();
 
   int rank() const { return rank_; }
   int world() const { return world_; }

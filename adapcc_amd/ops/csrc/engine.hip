@@ -292,6 +292,14 @@ void Engine::allreduce(void* data, long total_elems, int dtype, int op,
   HIP_CHECK(hipStreamWaitEvent(caller, ev_barrier_, 0));
 }
 
+std::string Engine::dump_inbox() {
+  std::string out(sizeof(FlagInbox), '\0');
+  HIP_CHECK(hipDeviceSynchronize());
+  HIP_CHECK(hipMemcpy(out.data(), tabs_.inbox[rank_], sizeof(FlagInbox),
+                      hipMemcpyDeviceToHost));
+  return out;
+}
+
 std::pair<uint64_t, uint64_t> Engine::query_error() {
   if (world_ == 1 || !connected_) return {0, 0};
   FlagInbox* inbox = tabs_.inbox[rank_];

@@ -26,7 +26,10 @@ static py::dict plan_to_dict(const adapcc::PlanData& p) {
     py::dict d;
     d["tree"] = u.tree; d["chunk"] = u.chunk;
     d["offset"] = u.offset_elems; d["count"] = u.count_elems;
-    d["notify"] = (bool)u.notify_parent; d["consumer"] = u.parent_rank;
+    d["flag_space"] = (int)u.flag_space;
+    py::list nl;
+    for (int i = 0; i < u.nnotify; ++i) nl.append(u.notify_rank[i]);
+    d["notify_to"] = nl;
     cs.append(d);
   }
   for (const auto& u : p.runits) {
@@ -48,7 +51,8 @@ static py::dict plan_to_dict(const adapcc::PlanData& p) {
   for (const auto& u : p.bunits) {
     py::dict d;
     d["tree"] = u.tree; d["chunk"] = u.chunk;
-    d["offset"] = u.offset_elems; d["count"] = u.count_elems;
+    d["src_offset"] = u.src_offset_elems; d["dst_offset"] = u.dst_offset_elems;
+    d["count"] = u.count_elems;
     d["parent"] = u.parent_rank; d["parent_kind"] = (int)u.parent_kind;
     d["forward"] = (bool)u.forward;
     py::list kids;
@@ -113,6 +117,42 @@ PYBIND11_MODULE(_core, m) {
         py::arg("esize"), py::arg("chunk_bytes"),
         py::arg("active") = std::vector<int>{});
 
+  m.def("compute_primitive_plan",
+        [](const std::string& prim, int world, int rank, long elems, int esize,
+           long chunk_bytes, int root,
+           const std::vector<std::vector<int>>& parents,
+           const std::vector<int>& active) {
+          uint64_t mask = 0;
+          for (int r : active) mask |= (1ull << r);
+          if (active.empty()) mask = (1ull << world) - 1;
+          adapcc::PlanData pd;
+          if (prim == "reduce") {
+            auto shape = parents.empty() ? adapcc::star_shape(world)
+                                         : adapcc::TreeShape::derive(parents);
+            pd = adapcc::build_reduce_plan(shape, rank, root, elems, esize,
+                                           chunk_bytes, mask);
+          } else if (prim == "broadcast") {
+            pd = adapcc::build_broadcast_plan(world, rank, root, elems, esize,
+                                              chunk_bytes);
+          } else if (prim == "allgather") {
+            pd = adapcc::build_allgather_plan(world, rank, elems, esize,
+                                              chunk_bytes);
+          } else if (prim == "reducescatter") {
+            pd = adapcc::build_reducescatter_plan(world, rank, elems, esize,
+                                                  chunk_bytes, mask);
+          } else if (prim == "alltoall") {
+            pd = adapcc::build_alltoall_plan(world, rank, elems, esize,
+                                             chunk_bytes);
+          } else {
+            throw std::runtime_error("unknown primitive " + prim);
+          }
+          return plan_to_dict(pd);
+        },
+        py::arg("prim"), py::arg("world"), py::arg("rank"), py::arg("elems"),
+        py::arg("esize"), py::arg("chunk_bytes"), py::arg("root") = 0,
+        py::arg("parents") = std::vector<std::vector<int>>{},
+        py::arg("active") = std::vector<int>{});
+
   py::class_<Engine>(m, "Engine")
       .def(py::init<int, int, int, size_t, double>(), py::arg("rank"),
            py::arg("world"), py::arg("device"), py::arg("cap_bytes"),
@@ -140,6 +180,53 @@ PYBIND11_MODULE(_core, m) {
            py::arg("data_ptr"), py::arg("numel"), py::arg("dtype"),
            py::arg("op"), py::arg("active"), py::arg("average"),
            py::arg("stream"))
+      .def("reduce",
+           [](Engine& e, uintptr_t data_ptr, long numel, int dtype, int op,
+              int root, const std::vector<int>& active, uintptr_t stream) {
+             e.reduce(reinterpret_cast<void*>(data_ptr), numel, dtype, op,
+                      root, active, reinterpret_cast<void*>(stream));
+           },
+           py::arg("data_ptr"), py::arg("numel"), py::arg("dtype"),
+           py::arg("op"), py::arg("root"), py::arg("active"),
+           py::arg("stream"))
+      .def("broadcast",
+           [](Engine& e, uintptr_t data_ptr, long numel, int dtype, int root,
+              uintptr_t stream) {
+             e.broadcast(reinterpret_cast<void*>(data_ptr), numel, dtype, root,
+                         reinterpret_cast<void*>(stream));
+           },
+           py::arg("data_ptr"), py::arg("numel"), py::arg("dtype"),
+           py::arg("root"), py::arg("stream"))
+      .def("all_gather",
+           [](Engine& e, uintptr_t in_ptr, uintptr_t out_ptr, long in_elems,
+              int dtype, uintptr_t stream) {
+             e.all_gather(reinterpret_cast<const void*>(in_ptr),
+                          reinterpret_cast<void*>(out_ptr), in_elems, dtype,
+                          reinterpret_cast<void*>(stream));
+           },
+           py::arg("in_ptr"), py::arg("out_ptr"), py::arg("in_elems"),
+           py::arg("dtype"), py::arg("stream"))
+      .def("all_to_all",
+           [](Engine& e, uintptr_t in_ptr, uintptr_t out_ptr, long per_rank,
+              int dtype, uintptr_t stream) {
+             e.all_to_all(reinterpret_cast<const void*>(in_ptr),
+                          reinterpret_cast<void*>(out_ptr), per_rank, dtype,
+                          reinterpret_cast<void*>(stream));
+           },
+           py::arg("in_ptr"), py::arg("out_ptr"), py::arg("per_rank_elems"),
+           py::arg("dtype"), py::arg("stream"))
+      .def("reduce_scatter",
+           [](Engine& e, uintptr_t in_ptr, uintptr_t out_ptr, long out_elems,
+              int dtype, int op, const std::vector<int>& active, bool average,
+              uintptr_t stream) {
+             e.reduce_scatter(reinterpret_cast<const void*>(in_ptr),
+                              reinterpret_cast<void*>(out_ptr), out_elems,
+                              dtype, op, active, average,
+                              reinterpret_cast<void*>(stream));
+           },
+           py::arg("in_ptr"), py::arg("out_ptr"), py::arg("out_elems"),
+           py::arg("dtype"), py::arg("op"), py::arg("active"),
+           py::arg("average"), py::arg("stream"))
       .def("synchronize", &Engine::synchronize,
            py::call_guard<py::gil_scoped_release>())
       .def("query_error", &Engine::query_error)

@@ -72,27 +72,34 @@ struct ReduceUnit {
   int child_rank[kMaxRanks];
 };
 
-// One copy-in unit: stage user-tensor range (t, chunk) into sendbuf and
-// optionally notify the tree parent that it is pullable.
+// One copy-in unit: stage a user-tensor range into sendbuf at (t, chunk)
+// and notify its consumers that it is pullable. flag_space selects which
+// inbox array the notification lands in: 0 -> ready[me][t][c] (a reducer
+// will pull), 1 -> bcast[t][c] (a receiver will pull, e.g. broadcast /
+// allgather source).
 struct CopyUnit {
   int tree;
   int chunk;
   long offset_elems;
   long count_elems;
-  uint8_t notify_parent;
-  int parent_rank;
+  uint8_t flag_space;  // 0 = ready, 1 = bcast
+  int nnotify;
+  int notify_rank[kMaxRanks];
 };
 
-// One broadcast/pull unit: wait for tree-t parent's publication of (t,chunk),
-// pull it into the user tensor (and forward: re-publish to own children).
+// One receive/pull unit: wait for the (t,chunk) publication in the local
+// bcast inbox, pull count elems from the source buffer at src_offset into
+// the user OUT tensor at dst_offset (scaled); forward: also write resultbuf
+// at src_offset and re-publish to children (multi-level trees).
 struct BcastUnit {
   int tree;
   int chunk;
-  long offset_elems;
+  long src_offset_elems;
+  long dst_offset_elems;
   long count_elems;
-  int parent_rank;      // who to pull from (-1: self is root -> local accbuf)
-  uint8_t parent_kind;  // BufKind of parent's published buffer
-  uint8_t forward;      // also write resultbuf + notify children
+  int parent_rank;      // who to pull from (-1: self -> local buffer)
+  uint8_t parent_kind;  // BufKind of the source buffer
+  uint8_t forward;
   int nchildren;
   int child_rank[kMaxRanks];
 };

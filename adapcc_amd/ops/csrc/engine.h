@@ -31,12 +31,22 @@ class Engine {
   void set_strategy(const std::vector<std::vector<int>>& parents,
                     long chunk_bytes);
 
-  // Enqueue an allreduce of `total_elems` elements of `dtype` at `data` on
-  // the caller stream. active_ranks empty => all ranks. average => divide
-  // by |active| on the broadcast write.
+  // Enqueue collectives on the caller stream (asynchronous; the caller
+  // stream is made to depend on completion). active_ranks empty => all.
   void allreduce(void* data, long total_elems, int dtype, int op,
                  const std::vector<int>& active_ranks, bool average,
                  void* caller_stream);
+  void reduce(void* data, long total_elems, int dtype, int op, int root,
+              const std::vector<int>& active_ranks, void* caller_stream);
+  void broadcast(void* data, long total_elems, int dtype, int root,
+                 void* caller_stream);
+  void all_gather(const void* in, void* out, long in_elems, int dtype,
+                  void* caller_stream);
+  void all_to_all(const void* in, void* out, long per_rank_elems, int dtype,
+                  void* caller_stream);
+  void reduce_scatter(const void* in, void* out, long out_elems, int dtype,
+                      int op, const std::vector<int>& active_ranks,
+                      bool average, void* caller_stream);
 
   void synchronize();
   std::pair<uint64_t, uint64_t> query_error();
@@ -63,17 +73,26 @@ class Engine {
 
  private:
   struct PlanKey {
+    int prim;
     long elems;
     int dt;
     int op;
     uint64_t mask;
+    int root;
     bool operator<(const PlanKey& o) const {
-      return std::tie(elems, dt, op, mask) < std::tie(o.elems, o.dt, o.op, o.mask);
+      return std::tie(prim, elems, dt, op, mask, root) <
+             std::tie(o.prim, o.elems, o.dt, o.op, o.mask, o.root);
     }
   };
 
   void build_tables();
-  Plan& get_plan(long total_elems, Dtype dt, RedOp op, uint64_t active_mask);
+  Plan& get_plan(int prim, long elems, Dtype dt, RedOp op, uint64_t active_mask,
+                 int root);
+  uint64_t resolve_mask(const std::vector<int>& active_ranks) const;
+  void enqueue(const Plan& plan, const void* in, void* out, CallArgs& args,
+               void* caller_stream);
+  CallArgs make_args(Dtype dt, RedOp op, float scale, long elems);
+  void check_ready(long bytes_needed) const;
 
   int rank_, world_, device_;
   size_t cap_bytes_;

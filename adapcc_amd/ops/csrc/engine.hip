@@ -179,25 +179,47 @@ void Engine::Plan::free_device() {
   d_c = nullptr; d_r = nullptr; d_b = nullptr; d_ranks = nullptr;
 }
 
-// Build the per-call unit plan (plan.cpp; deterministic across ranks).
-Engine::Plan& Engine::get_plan(long total_elems, Dtype dt, RedOp op,
-                               uint64_t active_mask) {
-  PlanKey key{total_elems, (int)dt, (int)op, active_mask};
+// Build or fetch the cached per-call unit plan (plan.cpp builders;
+// deterministic across ranks for a given key).
+Engine::Plan& Engine::get_plan(int prim, long elems, Dtype dt, RedOp op,
+                               uint64_t active_mask, int root) {
+  PlanKey key{prim, elems, (int)dt, (int)op, active_mask, root};
   auto it = plans_.find(key);
   if (it != plans_.end()) return it->second;
 
-  PlanData pd = build_plan(shape_, rank_, total_elems, dtype_size(dt),
-                           chunk_bytes_, active_mask);
+  const int esize = dtype_size(dt);
+  PlanData pd;
+  switch (prim) {
+    case 0:  // allreduce
+      pd = build_plan(shape_, rank_, elems, esize, chunk_bytes_, active_mask);
+      break;
+    case 1:  // reduce
+      pd = build_reduce_plan(shape_, rank_, root, elems, esize, chunk_bytes_,
+                             active_mask);
+      break;
+    case 2:  // broadcast
+      pd = build_broadcast_plan(world_, rank_, root, elems, esize, chunk_bytes_);
+      break;
+    case 3:  // allgather (elems = per-rank in elems)
+      pd = build_allgather_plan(world_, rank_, elems, esize, chunk_bytes_);
+      break;
+    case 4:  // alltoall (elems = per-rank slice elems)
+      pd = build_alltoall_plan(world_, rank_, elems, esize, chunk_bytes_);
+      break;
+    case 5:  // reducescatter (elems = out elems)
+      pd = build_reducescatter_plan(world_, rank_, elems, esize, chunk_bytes_,
+                                    active_mask);
+      break;
+    default:
+      throw std::runtime_error("unknown primitive");
+  }
+
   Plan plan;
-  plan.total_elems = total_elems;
+  plan.total_elems = elems;
   plan.dt = dt;
   plan.cunits = std::move(pd.cunits);
   plan.runits = std::move(pd.runits);
   plan.bunits = std::move(pd.bunits);
-
-  if ((int)plan.cunits.size() > kMaxUnits || (int)plan.runits.size() > kMaxUnits ||
-      (int)plan.bunits.size() > kMaxUnits)
-    throw std::runtime_error("too many units");
 
   // participating ranks for the end barrier: everyone
   std::vector<int> ranks(world_);
@@ -223,48 +245,26 @@ Engine::Plan& Engine::get_plan(long total_elems, Dtype dt, RedOp op,
   return res.first->second;
 }
 
-void Engine::allreduce(void* data, long total_elems, int dtype, int op,
-                       const std::vector<int>& active_ranks, bool average,
-                       void* caller_stream) {
-  if (!connected_ && world_ > 1)
-    throw std::runtime_error("engine not connected");
-  if (world_ == 1) {
-    return;  // sum over {self} is the identity; avg likewise
-  }
-  Dtype dt = (Dtype)dtype;
-  RedOp rop = (RedOp)op;
-  const long nbytes = total_elems * dtype_size(dt);
-  if ((size_t)nbytes > cap_bytes_)
-    throw std::runtime_error("tensor larger than engine capacity; split the call");
-
+uint64_t Engine::resolve_mask(const std::vector<int>& active_ranks) const {
   uint64_t mask = 0;
   if (active_ranks.empty()) {
     mask = (world_ >= 64) ? ~0ull : ((1ull << world_) - 1);
   } else {
     for (int r : active_ranks) mask |= (1ull << r);
   }
-  int n_active = __builtin_popcountll(mask);
-  if (n_active == 0) throw std::runtime_error("empty active set");
+  if (mask == 0) throw std::runtime_error("empty active set");
+  return mask;
+}
 
-  Plan& plan = get_plan(total_elems, dt, rop, mask);
-
-  CallArgs args{};
-  args.seq = ++seq_;
-  args.dtype = dt;
-  args.op = rop;
-  args.scale = (rop == RedOp::Avg) ? 1.0f / n_active : 1.0f;
-  args.total_elems = total_elems;
-  args.timeout_ticks = (uint64_t)(timeout_ms_ * 100000.0);
-
+void Engine::enqueue(const Plan& plan, const void* in, void* out,
+                     CallArgs& args, void* caller_stream) {
   hipStream_t caller = reinterpret_cast<hipStream_t>(caller_stream);
-
   HIP_CHECK(hipSetDevice(device_));
   // serialize after previous call + after caller-produced data
-  if (seq_ > 1) HIP_CHECK(hipStreamWaitEvent(s_red_, ev_barrier_, 0));
+  if (args.seq > 1) HIP_CHECK(hipStreamWaitEvent(s_red_, ev_barrier_, 0));
   HIP_CHECK(hipEventRecord(ev_in_, caller));
   HIP_CHECK(hipStreamWaitEvent(s_red_, ev_in_, 0));
 
-  // zero per-call unit counters
   const size_t cu64 = sizeof(unsigned long long);
   if (!plan.cunits.empty())
     HIP_CHECK(hipMemsetAsync(counters_, 0, plan.cunits.size() * cu64, s_red_));
@@ -277,19 +277,131 @@ void Engine::allreduce(void* data, long total_elems, int dtype, int op,
   HIP_CHECK(hipEventRecord(ev_sync0_, s_red_));
   HIP_CHECK(hipStreamWaitEvent(s_bcast_, ev_sync0_, 0));
 
-  launch_collective(dt, data, data, plan.d_c, (int)plan.cunits.size(), plan.d_r,
-                    (int)plan.runits.size(), plan.d_b, (int)plan.bunits.size(),
-                    tabs_, args, rank_, counters_ + kMaxUnits,
-                    counters_ + 2 * kMaxUnits, wgs_per_group_, n_groups_, s_red_,
-                    s_bcast_);
+  launch_collective(args.dtype, in, out, plan.d_c, (int)plan.cunits.size(),
+                    plan.d_r, (int)plan.runits.size(), plan.d_b,
+                    (int)plan.bunits.size(), tabs_, args, rank_,
+                    counters_ + kMaxUnits, counters_ + 2 * kMaxUnits,
+                    wgs_per_group_, n_groups_, s_red_, s_bcast_);
   HIP_CHECK(hipGetLastError());
 
   HIP_CHECK(hipEventRecord(ev_red_, s_red_));
   HIP_CHECK(hipStreamWaitEvent(s_bcast_, ev_red_, 0));
-  launch_barrier(tabs_, args, rank_, world_, plan.d_ranks, plan.nranks, s_bcast_);
+  launch_barrier(tabs_, args, rank_, world_, plan.d_ranks, plan.nranks,
+                 s_bcast_);
   HIP_CHECK(hipGetLastError());
   HIP_CHECK(hipEventRecord(ev_barrier_, s_bcast_));
   HIP_CHECK(hipStreamWaitEvent(caller, ev_barrier_, 0));
+}
+
+CallArgs Engine::make_args(Dtype dt, RedOp op, float scale, long elems) {
+  CallArgs args{};
+  args.seq = ++seq_;
+  args.dtype = dt;
+  args.op = op;
+  args.scale = scale;
+  args.total_elems = elems;
+  args.timeout_ticks = (uint64_t)(timeout_ms_ * 100000.0);
+  return args;
+}
+
+void Engine::check_ready(long bytes_needed) const {
+  if (!connected_ && world_ > 1)
+    throw std::runtime_error("engine not connected");
+  if ((size_t)bytes_needed > cap_bytes_)
+    throw std::runtime_error("tensor larger than engine capacity; split the call");
+}
+
+void Engine::allreduce(void* data, long total_elems, int dtype, int op,
+                       const std::vector<int>& active_ranks, bool average,
+                       void* caller_stream) {
+  if (world_ == 1) return;
+  Dtype dt = (Dtype)dtype;
+  RedOp rop = (RedOp)op;
+  check_ready(total_elems * dtype_size(dt));
+  uint64_t mask = resolve_mask(active_ranks);
+  int n_active = __builtin_popcountll(mask);
+  Plan& plan = get_plan(0, total_elems, dt, rop, mask, -1);
+  CallArgs args = make_args(dt, rop, (rop == RedOp::Avg) ? 1.0f / n_active : 1.0f,
+                            total_elems);
+  enqueue(plan, data, data, args, caller_stream);
+}
+
+void Engine::reduce(void* data, long total_elems, int dtype, int op, int root,
+                    const std::vector<int>& active_ranks, void* caller_stream) {
+  if (world_ == 1) return;
+  Dtype dt = (Dtype)dtype;
+  RedOp rop = (RedOp)op;
+  check_ready(total_elems * dtype_size(dt));
+  uint64_t mask = resolve_mask(active_ranks);
+  int n_active = __builtin_popcountll(mask);
+  Plan& plan = get_plan(1, total_elems, dt, rop, mask, root);
+  CallArgs args = make_args(dt, rop, (rop == RedOp::Avg) ? 1.0f / n_active : 1.0f,
+                            total_elems);
+  enqueue(plan, data, data, args, caller_stream);
+}
+
+void Engine::broadcast(void* data, long total_elems, int dtype, int root,
+                       void* caller_stream) {
+  if (world_ == 1) return;
+  Dtype dt = (Dtype)dtype;
+  check_ready(total_elems * dtype_size(dt));
+  Plan& plan = get_plan(2, total_elems, dt, RedOp::Sum,
+                        (1ull << world_) - 1, root);
+  CallArgs args = make_args(dt, RedOp::Sum, 1.0f, total_elems);
+  enqueue(plan, data, data, args, caller_stream);
+}
+
+void Engine::all_gather(const void* in, void* out, long in_elems, int dtype,
+                        void* caller_stream) {
+  Dtype dt = (Dtype)dtype;
+  check_ready((long)world_ * in_elems * dtype_size(dt));
+  if (world_ == 1) {
+    HIP_CHECK(hipMemcpyAsync(out, in, in_elems * dtype_size(dt),
+                             hipMemcpyDeviceToDevice,
+                             reinterpret_cast<hipStream_t>(caller_stream)));
+    return;
+  }
+  Plan& plan = get_plan(3, in_elems, dt, RedOp::Sum, (1ull << world_) - 1, -1);
+  CallArgs args = make_args(dt, RedOp::Sum, 1.0f, in_elems);
+  enqueue(plan, in, out, args, caller_stream);
+}
+
+void Engine::all_to_all(const void* in, void* out, long per_rank_elems,
+                        int dtype, void* caller_stream) {
+  Dtype dt = (Dtype)dtype;
+  check_ready((long)world_ * per_rank_elems * dtype_size(dt));
+  if (world_ == 1) {
+    HIP_CHECK(hipMemcpyAsync(out, in, per_rank_elems * dtype_size(dt),
+                             hipMemcpyDeviceToDevice,
+                             reinterpret_cast<hipStream_t>(caller_stream)));
+    return;
+  }
+  Plan& plan = get_plan(4, per_rank_elems, dt, RedOp::Sum,
+                        (1ull << world_) - 1, -1);
+  CallArgs args = make_args(dt, RedOp::Sum, 1.0f, per_rank_elems);
+  enqueue(plan, in, out, args, caller_stream);
+}
+
+void Engine::reduce_scatter(const void* in, void* out, long out_elems,
+                            int dtype, int op,
+                            const std::vector<int>& active_ranks, bool average,
+                            void* caller_stream) {
+  Dtype dt = (Dtype)dtype;
+  RedOp rop = (RedOp)op;
+  check_ready((long)world_ * out_elems * dtype_size(dt));
+  if (world_ == 1) {
+    HIP_CHECK(hipMemcpyAsync(out, in, out_elems * dtype_size(dt),
+                             hipMemcpyDeviceToDevice,
+                             reinterpret_cast<hipStream_t>(caller_stream)));
+    return;
+  }
+  uint64_t mask = resolve_mask(active_ranks);
+  int n_active = __builtin_popcountll(mask);
+  Plan& plan = get_plan(5, out_elems, dt, rop, mask, -1);
+  CallArgs args = make_args(dt, rop,
+                            (average || rop == RedOp::Avg) ? 1.0f / n_active : 1.0f,
+                            out_elems);
+  enqueue(plan, in, out, args, caller_stream);
 }
 
 std::string Engine::dump_inbox() {

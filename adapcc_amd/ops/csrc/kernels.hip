@@ -210,10 +210,15 @@ __global__ void __launch_bounds__(256) copyin_kernel(
       dst[j] = src[j];
     }
 
-    if (u.notify_parent) {
+    if (u.nnotify > 0) {
       const bool last = unit_arrive(&tabs.counters[ui], wgs_per_group);
       if (last && threadIdx.x == 0) {
-        push_flag(&tabs.inbox[u.parent_rank]->ready[me][u.tree][u.chunk], args.seq);
+        for (int c = 0; c < u.nnotify; ++c) {
+          FlagInbox* ib = tabs.inbox[u.notify_rank[c]];
+          uint64_t* f = (u.flag_space == 0) ? &ib->ready[me][u.tree][u.chunk]
+                                            : &ib->bcast[u.tree][u.chunk];
+          push_flag(f, args.seq);
+        }
       }
     }
   }
@@ -351,12 +356,14 @@ __global__ void __launch_bounds__(256) bcast_kernel(
       return;
     }
     const int src_rank = u.parent_rank < 0 ? me : u.parent_rank;
-    const void* base = (BufKind)u.parent_kind == BufKind::Acc
-                           ? (const void*)tabs.acc[src_rank]
-                           : (const void*)tabs.result[src_rank];
-    const T* src = (const T*)base + u.offset_elems;
-    T* dst = user + u.offset_elems;
-    T* fwd = u.forward ? (T*)tabs.result[me] + u.offset_elems : nullptr;
+    const void* base = (BufKind)u.parent_kind == BufKind::Send
+                           ? tabs.send[src_rank]
+                           : (BufKind)u.parent_kind == BufKind::Acc
+                                 ? (const void*)tabs.acc[src_rank]
+                                 : (const void*)tabs.result[src_rank];
+    const T* src = (const T*)base + u.src_offset_elems;
+    T* dst = user + u.dst_offset_elems;
+    T* fwd = u.forward ? (T*)tabs.result[me] + u.src_offset_elems : nullptr;
 
     const long cnt = u.count_elems;
     const long per =

@@ -1,4 +1,17 @@
-// See plan.h. Extracted from the engine so CPU tests can drive it.
+// Host-only unit-plan builders for every primitive (see plan.h).
+//
+// All plans are deterministic functions of (shape/world, sizes, active mask)
+// so every rank independently derives the same flag grid. Single-node direct
+// algorithms (every pair of MI355X GPUs has a dedicated xGMI link):
+//   allreduce      tree forest: chunked reduce phase + broadcast phase
+//   reduce         forest reduce phase + per-slice gather at the root
+//   broadcast      receivers pull root's staged slices directly
+//   allgather      receivers pull each source's staged tensor directly
+//   reducescatter  star forest reduce phase, slice t owned by rank t
+//   alltoall       (src,dst)-tiled direct pulls from the source's sendbuf
+// Reference parity: these were declared but unimplemented upstream
+// (trans.h:31-33); reduce/broadcast existed as tree contexts (reduce.cu,
+// boardcast.cu).
 
 #include "plan.h"
 
@@ -31,7 +44,6 @@ TreeShape TreeShape::derive(const std::vector<std::vector<int>>& parents) {
       }
     }
     if (s.roots[t] < 0) throw std::runtime_error("tree has no root");
-    // cycle check: every rank must reach the root
     for (int r = 0; r < s.world; ++r) {
       int cur = r, hops = 0;
       while (cur != s.roots[t]) {
@@ -44,7 +56,16 @@ TreeShape TreeShape::derive(const std::vector<std::vector<int>>& parents) {
   return s;
 }
 
+TreeShape star_shape(int world) {
+  std::vector<std::vector<int>> parents(world, std::vector<int>(world));
+  for (int t = 0; t < world; ++t)
+    for (int r = 0; r < world; ++r) parents[t][r] = (r == t) ? -1 : t;
+  return TreeShape::derive(parents);
+}
+
 namespace {
+
+constexpr long kAlignE = 64;
 
 struct Provider {
   int rank;
@@ -62,6 +83,10 @@ struct TreeAnalysis {
   int root = -1;
 };
 
+// Relay / effective-source analysis (reference: csrc/control.cu:27-101):
+// inactive node + single active inflow -> skipped (consumer pulls the inflow
+// directly over xGMI); >=2 inflows -> still aggregates; the root always
+// materializes its tree's result.
 std::vector<Provider> analyze_subtree(int node,
                                       const std::vector<std::vector<int>>& children,
                                       const std::vector<char>& active,
@@ -93,52 +118,62 @@ std::vector<Provider> analyze_subtree(int node,
   return below;
 }
 
-}  // namespace
-
-PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
-                    int esize, long chunk_bytes, uint64_t active_mask) {
-  const int world = shape.world;
-  const int T = (int)shape.parents.size();
+std::vector<char> mask_to_active(uint64_t mask, int world) {
   std::vector<char> active(world, 0);
   for (int r = 0; r < world; ++r)
-    if (active_mask & (1ull << r)) active[r] = 1;
+    if (mask & (1ull << r)) active[r] = 1;
+  return active;
+}
 
-  const long align_e = 64;
-  const long per_raw = (total_elems + T - 1) / T;
-  const long per = ((per_raw + align_e - 1) / align_e) * align_e;
-  long chunk_elems = std::max<long>(chunk_bytes / esize, align_e);
-  chunk_elems = ((chunk_elems + align_e - 1) / align_e) * align_e;
-  while ((per + chunk_elems - 1) / chunk_elems > kMaxChunkSlots) chunk_elems *= 2;
+long pick_chunk_elems(long slice_elems, int esize, long chunk_bytes,
+                      int max_slots = kMaxChunkSlots) {
+  long ce = std::max<long>(chunk_bytes / esize, kAlignE);
+  ce = (ce + kAlignE - 1) / kAlignE * kAlignE;
+  while ((slice_elems + ce - 1) / ce > max_slots) ce *= 2;
+  return ce;
+}
 
-  PlanData plan;
-  plan.chunk_elems = chunk_elems;
+struct TC {
+  int t;
+  int c;
+  long off;
+  long cnt;
+};
 
-  std::vector<TreeAnalysis> ana(T);
-  for (int t = 0; t < T; ++t) {
-    ana[t].root = shape.roots[t];
-    analyze_subtree(shape.roots[t], shape.children[t], active, ana[t]);
-  }
-
-  struct TC { int t; int c; long off; long cnt; };
+// (chunk, tree)-ordered grid over per-tree slices [beg, end).
+std::vector<TC> make_grid(const std::vector<std::pair<long, long>>& slice,
+                          long chunk_elems) {
   std::vector<TC> grid;
   long max_chunks = 0;
-  std::vector<std::pair<long, long>> slice(T);
-  for (int t = 0; t < T; ++t) {
-    long beg = std::min((long)t * per, total_elems);
-    long end = std::min(beg + per, total_elems);
-    slice[t] = {beg, end};
-    long n = (end - beg + chunk_elems - 1) / chunk_elems;
+  for (const auto& s : slice) {
+    long n = (s.second - s.first + chunk_elems - 1) / chunk_elems;
     max_chunks = std::max(max_chunks, n);
   }
   for (long c = 0; c < max_chunks; ++c) {
-    for (int t = 0; t < T; ++t) {
+    for (int t = 0; t < (int)slice.size(); ++t) {
       long beg = slice[t].first + c * chunk_elems;
       if (beg >= slice[t].second) continue;
-      long cnt = std::min(chunk_elems, slice[t].second - beg);
-      grid.push_back({t, (int)c, beg, cnt});
+      grid.push_back({t, (int)c, beg, std::min(chunk_elems, slice[t].second - beg)});
     }
   }
+  return grid;
+}
 
+void check_units(const PlanData& p) {
+  const long maxu = (long)kMaxTrees * kMaxChunkSlots;
+  if ((long)p.cunits.size() > maxu || (long)p.runits.size() > maxu ||
+      (long)p.bunits.size() > maxu)
+    throw std::runtime_error("too many units");
+}
+
+// Shared forest reduce phase (copy-in + reduce units); publishes each tree's
+// materialized root result to `publish_to(tree_root)`.
+void forest_reduce_phase(const TreeShape& shape, int rank,
+                         const std::vector<TC>& grid,
+                         const std::vector<TreeAnalysis>& ana,
+                         const std::vector<char>& active,
+                         const std::vector<std::vector<int>>& publish_to,
+                         PlanData& plan) {
   const bool me_active = active[rank] != 0;
   for (const auto& tc : grid) {
     const TreeAnalysis& A = ana[tc.t];
@@ -148,10 +183,11 @@ PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
       cu.chunk = tc.c;
       cu.offset_elems = tc.off;
       cu.count_elems = tc.cnt;
+      cu.flag_space = 0;
       auto sc = A.send_consumer.find(rank);
       if (sc != A.send_consumer.end()) {
-        cu.notify_parent = 1;
-        cu.parent_rank = sc->second;
+        cu.nnotify = 1;
+        cu.notify_rank[0] = sc->second;
       }
       plan.cunits.push_back(cu);
     }
@@ -175,34 +211,284 @@ PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
       }
       if (rank == A.root) {
         ru.is_root = 1;
-        ru.child_rank[ru.nchildren++] = rank;
-        for (int c : shape.children[tc.t][rank])
-          ru.child_rank[ru.nchildren++] = c;
+        for (int k : publish_to[tc.t]) ru.child_rank[ru.nchildren++] = k;
       }
       plan.runits.push_back(ru);
     }
-    {
+  }
+}
+
+}  // namespace
+
+PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
+                    int esize, long chunk_bytes, uint64_t active_mask) {
+  const int world = shape.world;
+  const int T = (int)shape.parents.size();
+  auto active = mask_to_active(active_mask, world);
+
+  const long per_raw = (total_elems + T - 1) / T;
+  const long per = (per_raw + kAlignE - 1) / kAlignE * kAlignE;
+  const long chunk_elems = pick_chunk_elems(per, esize, chunk_bytes);
+
+  PlanData plan;
+  plan.chunk_elems = chunk_elems;
+
+  std::vector<TreeAnalysis> ana(T);
+  std::vector<std::vector<int>> publish(T);
+  for (int t = 0; t < T; ++t) {
+    ana[t].root = shape.roots[t];
+    analyze_subtree(shape.roots[t], shape.children[t], active, ana[t]);
+    // allreduce: root publishes to itself + its direct children
+    publish[t].push_back(shape.roots[t]);
+    for (int c : shape.children[t][shape.roots[t]]) publish[t].push_back(c);
+  }
+
+  std::vector<std::pair<long, long>> slice(T);
+  for (int t = 0; t < T; ++t) {
+    slice[t] = {std::min((long)t * per, total_elems),
+                std::min((long)(t + 1) * per, total_elems)};
+  }
+  auto grid = make_grid(slice, chunk_elems);
+  forest_reduce_phase(shape, rank, grid, ana, active, publish, plan);
+
+  // broadcast phase: every rank receives every chunk
+  for (const auto& tc : grid) {
+    const TreeAnalysis& A = ana[tc.t];
+    BcastUnit bu{};
+    bu.tree = tc.t;
+    bu.chunk = tc.c;
+    bu.src_offset_elems = tc.off;
+    bu.dst_offset_elems = tc.off;
+    bu.count_elems = tc.cnt;
+    if (rank == A.root) {
+      bu.parent_rank = -1;
+      bu.parent_kind = (uint8_t)BufKind::Acc;
+    } else {
+      int p = shape.parents[tc.t][rank];
+      bu.parent_rank = p;
+      bu.parent_kind = (uint8_t)(p == A.root ? BufKind::Acc : BufKind::Result);
+    }
+    const auto& kids = shape.children[tc.t][rank];
+    if (rank != A.root && !kids.empty()) {
+      bu.forward = 1;
+      for (int c : kids) bu.child_rank[bu.nchildren++] = c;
+    }
+    plan.bunits.push_back(bu);
+  }
+  check_units(plan);
+  return plan;
+}
+
+PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
+                           long total_elems, int esize, long chunk_bytes,
+                           uint64_t active_mask) {
+  const int world = shape.world;
+  const int T = (int)shape.parents.size();
+  auto active = mask_to_active(active_mask, world);
+
+  const long per_raw = (total_elems + T - 1) / T;
+  const long per = (per_raw + kAlignE - 1) / kAlignE * kAlignE;
+  const long chunk_elems = pick_chunk_elems(per, esize, chunk_bytes);
+
+  PlanData plan;
+  plan.chunk_elems = chunk_elems;
+
+  std::vector<TreeAnalysis> ana(T);
+  std::vector<std::vector<int>> publish(T);
+  for (int t = 0; t < T; ++t) {
+    ana[t].root = shape.roots[t];
+    analyze_subtree(shape.roots[t], shape.children[t], active, ana[t]);
+    publish[t] = {root};  // each tree's result goes to THE root only
+  }
+  std::vector<std::pair<long, long>> slice(T);
+  for (int t = 0; t < T; ++t) {
+    slice[t] = {std::min((long)t * per, total_elems),
+                std::min((long)(t + 1) * per, total_elems)};
+  }
+  auto grid = make_grid(slice, chunk_elems);
+  forest_reduce_phase(shape, rank, grid, ana, active, publish, plan);
+
+  if (rank == root) {
+    for (const auto& tc : grid) {
       BcastUnit bu{};
       bu.tree = tc.t;
       bu.chunk = tc.c;
-      bu.offset_elems = tc.off;
+      bu.src_offset_elems = tc.off;
+      bu.dst_offset_elems = tc.off;
       bu.count_elems = tc.cnt;
-      if (rank == A.root) {
-        bu.parent_rank = -1;
-        bu.parent_kind = (uint8_t)BufKind::Acc;
-      } else {
-        int p = shape.parents[tc.t][rank];
-        bu.parent_rank = p;
-        bu.parent_kind = (uint8_t)(p == A.root ? BufKind::Acc : BufKind::Result);
-      }
-      const auto& kids = shape.children[tc.t][rank];
-      if (rank != A.root && !kids.empty()) {
-        bu.forward = 1;
-        for (int c : kids) bu.child_rank[bu.nchildren++] = c;
-      }
+      int troot = shape.roots[tc.t];
+      bu.parent_rank = (troot == rank) ? -1 : troot;
+      bu.parent_kind = (uint8_t)BufKind::Acc;
       plan.bunits.push_back(bu);
     }
   }
+  check_units(plan);
+  return plan;
+}
+
+PlanData build_broadcast_plan(int world, int rank, int root, long total_elems,
+                              int esize, long chunk_bytes) {
+  const int T = std::min<long>(std::min(world, 8), kMaxTrees);
+  const long per_raw = (total_elems + T - 1) / T;
+  const long per = (per_raw + kAlignE - 1) / kAlignE * kAlignE;
+  const long chunk_elems = pick_chunk_elems(per, esize, chunk_bytes);
+
+  PlanData plan;
+  plan.chunk_elems = chunk_elems;
+  std::vector<std::pair<long, long>> slice(T);
+  for (int t = 0; t < T; ++t) {
+    slice[t] = {std::min((long)t * per, total_elems),
+                std::min((long)(t + 1) * per, total_elems)};
+  }
+  auto grid = make_grid(slice, chunk_elems);
+  for (const auto& tc : grid) {
+    if (rank == root) {
+      CopyUnit cu{};
+      cu.tree = tc.t;
+      cu.chunk = tc.c;
+      cu.offset_elems = tc.off;
+      cu.count_elems = tc.cnt;
+      cu.flag_space = 1;  // receivers pull directly
+      for (int r = 0; r < world; ++r)
+        if (r != root) cu.notify_rank[cu.nnotify++] = r;
+      plan.cunits.push_back(cu);
+    } else {
+      BcastUnit bu{};
+      bu.tree = tc.t;
+      bu.chunk = tc.c;
+      bu.src_offset_elems = tc.off;
+      bu.dst_offset_elems = tc.off;
+      bu.count_elems = tc.cnt;
+      bu.parent_rank = root;
+      bu.parent_kind = (uint8_t)BufKind::Send;
+      plan.bunits.push_back(bu);
+    }
+  }
+  check_units(plan);
+  return plan;
+}
+
+PlanData build_allgather_plan(int world, int rank, long in_elems, int esize,
+                              long chunk_bytes) {
+  // tree s = source rank s; each source stages its in tensor; everyone
+  // pulls each source's chunks into out[s*L ..]
+  const long L = in_elems;
+  const long chunk_elems = pick_chunk_elems(L, esize, chunk_bytes);
+  PlanData plan;
+  plan.chunk_elems = chunk_elems;
+  const long nch = (L + chunk_elems - 1) / chunk_elems;
+  for (long c = 0; c < nch; ++c) {
+    const long off = c * chunk_elems;
+    const long cnt = std::min(chunk_elems, L - off);
+    for (int s = 0; s < world; ++s) {
+      if (s == rank) {
+        CopyUnit cu{};
+        cu.tree = s;
+        cu.chunk = (int)c;
+        cu.offset_elems = off;
+        cu.count_elems = cnt;
+        cu.flag_space = 1;
+        for (int r = 0; r < world; ++r) cu.notify_rank[cu.nnotify++] = r;
+        plan.cunits.push_back(cu);
+      }
+      BcastUnit bu{};
+      bu.tree = s;
+      bu.chunk = (int)c;
+      bu.src_offset_elems = off;
+      bu.dst_offset_elems = (long)s * L + off;
+      bu.count_elems = cnt;
+      bu.parent_rank = (s == rank) ? -1 : s;
+      bu.parent_kind = (uint8_t)BufKind::Send;
+      plan.bunits.push_back(bu);
+    }
+  }
+  check_units(plan);
+  return plan;
+}
+
+PlanData build_reducescatter_plan(int world, int rank, long out_elems,
+                                  int esize, long chunk_bytes,
+                                  uint64_t active_mask) {
+  // star forest with slice t = rank t's out range [t*L, (t+1)*L) of the
+  // world*L input
+  const long L = out_elems;
+  auto shape = star_shape(world);
+  auto active = mask_to_active(active_mask, world);
+  const long chunk_elems = pick_chunk_elems(L, esize, chunk_bytes);
+
+  PlanData plan;
+  plan.chunk_elems = chunk_elems;
+  std::vector<TreeAnalysis> ana(world);
+  std::vector<std::vector<int>> publish(world);
+  for (int t = 0; t < world; ++t) {
+    ana[t].root = t;
+    analyze_subtree(t, shape.children[t], active, ana[t]);
+    publish[t] = {t};
+  }
+  std::vector<std::pair<long, long>> slice(world);
+  for (int t = 0; t < world; ++t) slice[t] = {(long)t * L, (long)(t + 1) * L};
+  auto grid = make_grid(slice, chunk_elems);
+  forest_reduce_phase(shape, rank, grid, ana, active, publish, plan);
+
+  for (const auto& tc : grid) {
+    if (tc.t != rank) continue;
+    BcastUnit bu{};
+    bu.tree = tc.t;
+    bu.chunk = tc.c;
+    bu.src_offset_elems = tc.off;
+    bu.dst_offset_elems = tc.off - (long)rank * L;
+    bu.count_elems = tc.cnt;
+    bu.parent_rank = -1;
+    bu.parent_kind = (uint8_t)BufKind::Acc;
+    plan.bunits.push_back(bu);
+  }
+  check_units(plan);
+  return plan;
+}
+
+PlanData build_alltoall_plan(int world, int rank, long per_rank_elems,
+                             int esize, long chunk_bytes) {
+  // (src s, dst d) tiles: s stages its whole in; d pulls in[d*L..] of s into
+  // out[s*L..]. Flag slot = d*C + c (C chunks per L-slice), tree = s.
+  const long L = per_rank_elems;
+  const long chunk_elems =
+      pick_chunk_elems(L, esize, chunk_bytes, kMaxChunkSlots / world);
+  const long C = (L + chunk_elems - 1) / chunk_elems;
+  PlanData plan;
+  plan.chunk_elems = chunk_elems;
+  for (long c = 0; c < C; ++c) {
+    const long loff = c * chunk_elems;
+    const long cnt = std::min(chunk_elems, L - loff);
+    for (int d = 0; d < world; ++d) {
+      const int slot = (int)(d * C + c);
+      {  // my copy unit for destination d
+        CopyUnit cu{};
+        cu.tree = rank;
+        cu.chunk = slot;
+        cu.offset_elems = (long)d * L + loff;
+        cu.count_elems = cnt;
+        cu.flag_space = 1;
+        cu.nnotify = 1;
+        cu.notify_rank[0] = d;
+        plan.cunits.push_back(cu);
+      }
+      if (d == rank) {
+        // I receive from every source s the range [rank*L ..] of s's in
+        for (int s = 0; s < world; ++s) {
+          BcastUnit bu{};
+          bu.tree = s;
+          bu.chunk = slot;
+          bu.src_offset_elems = (long)rank * L + loff;
+          bu.dst_offset_elems = (long)s * L + loff;
+          bu.count_elems = cnt;
+          bu.parent_rank = (s == rank) ? -1 : s;
+          bu.parent_kind = (uint8_t)BufKind::Send;
+          plan.bunits.push_back(bu);
+        }
+      }
+    }
+  }
+  check_units(plan);
   return plan;
 }
 

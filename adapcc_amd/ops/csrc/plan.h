@@ -37,4 +37,21 @@ struct PlanData {
 PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
                     int esize, long chunk_bytes, uint64_t active_mask);
 
+// Remaining primitives (single-node direct algorithms; see plan.cpp).
+PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
+                           long total_elems, int esize, long chunk_bytes,
+                           uint64_t active_mask);
+PlanData build_broadcast_plan(int world, int rank, int root, long total_elems,
+                              int esize, long chunk_bytes);
+PlanData build_allgather_plan(int world, int rank, long in_elems, int esize,
+                              long chunk_bytes);
+PlanData build_reducescatter_plan(int world, int rank, long out_elems,
+                                  int esize, long chunk_bytes,
+                                  uint64_t active_mask);
+PlanData build_alltoall_plan(int world, int rank, long per_rank_elems,
+                             int esize, long chunk_bytes);
+
+// Fully-connected star forest (tree t rooted at rank t).
+TreeShape star_shape(int world);
+
 }  // namespace adapcc

@@ -150,6 +150,87 @@ class NativeEngine:
                                     active_list, average, stream)
         return tensor
 
+    def reduce(
+        self,
+        tensor: torch.Tensor,
+        root: int = 0,
+        active: Optional[Sequence[int]] = None,
+        average: bool = False,
+    ) -> torch.Tensor:
+        if self.world_size == 1:
+            return tensor
+        self._check(tensor)
+        core = _core()
+        op = core.OP_AVG if average else core.OP_SUM
+        self._eng.reduce(tensor.data_ptr(), tensor.numel(),
+                         _dtype_code(tensor.dtype), op, root,
+                         list(active) if active else [],
+                         self._stream(tensor))
+        return tensor
+
+    def broadcast(self, tensor: torch.Tensor, root: int = 0) -> torch.Tensor:
+        if self.world_size == 1:
+            return tensor
+        self._check(tensor)
+        self._eng.broadcast(tensor.data_ptr(), tensor.numel(),
+                            _dtype_code(tensor.dtype), root,
+                            self._stream(tensor))
+        return tensor
+
+    def all_gather(self, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        self._check(tensor)
+        self._check(out)
+        if out.numel() != tensor.numel() * self.world_size:
+            raise ValueError("all_gather: out must be world_size * in")
+        self._eng.all_gather(tensor.data_ptr(), out.data_ptr(), tensor.numel(),
+                             _dtype_code(tensor.dtype), self._stream(tensor))
+        return out
+
+    def all_to_all(self, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        self._check(tensor)
+        self._check(out)
+        if out.numel() != tensor.numel():
+            raise ValueError("all_to_all: out must match in size")
+        if tensor.numel() % self.world_size:
+            raise ValueError("all_to_all: size must divide world_size")
+        per = tensor.numel() // self.world_size
+        self._eng.all_to_all(tensor.data_ptr(), out.data_ptr(), per,
+                             _dtype_code(tensor.dtype), self._stream(tensor))
+        return out
+
+    def reduce_scatter(
+        self,
+        out: torch.Tensor,
+        tensor: torch.Tensor,
+        active: Optional[Sequence[int]] = None,
+        average: bool = False,
+    ) -> torch.Tensor:
+        self._check(tensor)
+        self._check(out)
+        if tensor.numel() != out.numel() * self.world_size:
+            raise ValueError("reduce_scatter: in must be world_size * out")
+        core = _core()
+        op = core.OP_AVG if average else core.OP_SUM
+        self._eng.reduce_scatter(tensor.data_ptr(), out.data_ptr(),
+                                 out.numel(), _dtype_code(tensor.dtype), op,
+                                 list(active) if active else [], average,
+                                 self._stream(tensor))
+        return out
+
+    def _check(self, tensor: torch.Tensor) -> None:
+        if not tensor.is_contiguous():
+            raise ValueError("adapcc collective requires a contiguous tensor")
+        if not tensor.is_cuda:
+            raise ValueError("adapcc native engine requires a GPU tensor")
+        if not self._connected:
+            raise RuntimeError("engine not bootstrapped")
+        if not self._strategy_set:
+            raise RuntimeError("no strategy set")
+
+    @staticmethod
+    def _stream(tensor: torch.Tensor) -> int:
+        return torch.cuda.current_stream(tensor.device).cuda_stream
+
     def synchronize(self) -> None:
         self._eng.synchronize()
 

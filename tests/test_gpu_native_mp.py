@@ -85,3 +85,126 @@ def test_native_average():
     assert all(run_mp(_world_case, 2, backend="gloo",
                       args=(123_457, [], True, "float32", 2),
                       timeout=300))
+
+
+def _prims_case(rank, world):
+    os.environ["ADAPCC_TIMEOUT_MS"] = "20000"
+    import torch
+    import torch.distributed as dist
+
+    torch.cuda.set_device(0)
+    from adapcc_amd.runtime.engine import NativeEngine
+    from adapcc_amd.strategy.partrees import synthesize_stars
+
+    eng = NativeEngine(rank, world, device=0, cap_bytes=64 << 20)
+    eng.bootstrap()
+    eng.set_strategy(synthesize_stars(world))
+
+    L = 100_000
+
+    def ref_gather(t):
+        cpu = t.float().cpu()
+        g = [torch.zeros_like(cpu) for _ in range(world)]
+        dist.all_gather(g, cpu)
+        return g
+
+    # reduce to root 1
+    torch.manual_seed(10 + rank)
+    t = torch.randn(L, device="cuda")
+    g = ref_gather(t)
+    eng.reduce(t, root=1 % world)
+    eng.synchronize()
+    if rank == 1 % world:
+        torch.testing.assert_close(t.cpu(), torch.stack(g).sum(0),
+                                   rtol=1e-4, atol=1e-4)
+
+    # broadcast from root 0
+    torch.manual_seed(20 + rank)
+    t = torch.randn(L, device="cuda")
+    g = ref_gather(t)
+    eng.broadcast(t, root=0)
+    eng.synchronize()
+    torch.testing.assert_close(t.cpu(), g[0], rtol=1e-6, atol=1e-6)
+
+    # allgather
+    torch.manual_seed(30 + rank)
+    t = torch.randn(L, device="cuda")
+    g = ref_gather(t)
+    out = torch.zeros(world * L, device="cuda")
+    eng.all_gather(out, t)
+    eng.synchronize()
+    torch.testing.assert_close(out.cpu(), torch.cat(g), rtol=1e-6, atol=1e-6)
+
+    # reduce_scatter
+    torch.manual_seed(40 + rank)
+    t = torch.randn(world * L, device="cuda")
+    g = ref_gather(t)
+    out = torch.zeros(L, device="cuda")
+    eng.reduce_scatter(out, t)
+    eng.synchronize()
+    total = torch.stack(g).sum(0)
+    torch.testing.assert_close(out.cpu(), total[rank * L:(rank + 1) * L],
+                               rtol=1e-4, atol=1e-4)
+
+    # alltoall
+    torch.manual_seed(50 + rank)
+    t = torch.randn(world * L, device="cuda")
+    g = ref_gather(t)
+    out = torch.zeros(world * L, device="cuda")
+    eng.all_to_all(out, t)
+    eng.synchronize()
+    expect = torch.cat([g[s][rank * L:(rank + 1) * L] for s in range(world)])
+    torch.testing.assert_close(out.cpu(), expect, rtol=1e-6, atol=1e-6)
+    return True
+
+
+@pytest.mark.parametrize("world", [2, 4])
+def test_native_other_primitives(world):
+    assert all(run_mp(_prims_case, world, backend="gloo", timeout=300))
+
+
+def _ddp_hook_gpu(rank, world):
+    os.environ["ADAPCC_TIMEOUT_MS"] = "20000"
+    os.environ["ADAPCC_TRANSPORT"] = "native"
+    import torch
+    import torch.nn as nn
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    torch.cuda.set_device(0)
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.runtime.hook import AdapccDDPState, adapcc_allreduce_hook
+
+    AdapCC.init(CommArgs(entry_point=-1), 0, rank, world)
+    AdapCC.setup()
+
+    def make_model():
+        torch.manual_seed(7)
+        return nn.Sequential(nn.Linear(64, 256), nn.GELU(),
+                             nn.Linear(256, 32)).cuda()
+
+    def data():
+        torch.manual_seed(900 + rank)
+        return (torch.randn(16, 64, device="cuda"),
+                torch.randn(16, 32, device="cuda"))
+
+    m1 = DDP(make_model(), bucket_cap_mb=1)
+    state = AdapccDDPState(AdapCC.communicator)
+    m1.register_comm_hook(state, adapcc_allreduce_hook)
+    x, y = data()
+    state.on_step(0)
+    ((m1(x) - y) ** 2).mean().backward()
+    torch.cuda.synchronize()
+    g1 = torch.cat([p.grad.flatten() for p in m1.parameters()])
+
+    m2 = DDP(make_model(), bucket_cap_mb=1)
+    ((m2(x) - y) ** 2).mean().backward()
+    torch.cuda.synchronize()
+    g2 = torch.cat([p.grad.flatten() for p in m2.parameters()])
+
+    assert torch.allclose(g1, g2, atol=1e-5), (g1 - g2).abs().max().item()
+    AdapCC.clear()
+    return True
+
+
+def test_ddp_hook_native_gpu():
+    assert all(run_mp(_ddp_hook_gpu, 2, backend="gloo", timeout=300))

@@ -1,11 +1,11 @@
 """Multi-rank protocol simulation over the native engine's unit plans.
 
-Drives the exact C++ plan/relay logic (adapcc_amd._core.compute_plan — the
-same code path Engine::get_plan uses on GPU) through a discrete-event
-simulation with numpy buffers: units execute only when their flag waits are
-satisfied, in worst-case serialized order (one group), so completion proves
-both numerical correctness and deadlock-freedom of the flag protocol for
-the given strategy/active-set.
+Drives the exact C++ plan/relay logic (adapcc_amd._core.compute_plan /
+compute_primitive_plan — the same code the Engine runs on GPU) through a
+discrete-event simulation with numpy buffers: units execute only when their
+flag waits are satisfied, in worst-case serialized order, so completion
+proves both numerical correctness and deadlock-freedom of the flag protocol
+for every primitive / strategy / active-set combination below.
 """
 
 import numpy as np
@@ -44,173 +44,184 @@ def binary_trees(n, ntrees=2):
     return out
 
 
-def simulate(parents, total, active=None, chunk_bytes=256, average=False):
-    """Returns (out_per_rank, n_iterations). Raises on deadlock."""
-    world = len(parents[0])
-    if active is None:
-        active = list(range(world))
-    plans = [
-        core.compute_plan(parents, r, total, 4, chunk_bytes, active)
-        for r in range(world)
-    ]
-    rng = np.random.default_rng(0)
-    user = [rng.standard_normal(total).astype(np.float32) for _ in range(world)]
-    out = [np.zeros(total, dtype=np.float32) for _ in range(world)]
-    send = [np.zeros(total, dtype=np.float32) for _ in range(world)]
-    acc = [np.zeros(total, dtype=np.float32) for _ in range(world)]
-    result = [np.zeros(total, dtype=np.float32) for _ in range(world)]
-    ready = set()   # (dst_rank, src_rank, tree, chunk)
-    bcast = set()   # (dst_rank, tree, chunk)
+class Sim:
+    """Execute per-rank unit plans against numpy buffers."""
 
-    scale = 1.0 / len(active) if average else 1.0
+    def __init__(self, world, plans, user_in, out_elems, scale=1.0):
+        self.world = world
+        self.plans = plans
+        self.user = user_in
+        total = max(len(u) for u in user_in)
+        self.out = [np.full(out_elems, np.nan, dtype=np.float32)
+                    for _ in range(world)]
+        self.send = [np.zeros(total, dtype=np.float32) for _ in range(world)]
+        self.acc = [np.zeros(total, dtype=np.float32) for _ in range(world)]
+        self.result = [np.zeros(total, dtype=np.float32) for _ in range(world)]
+        self.ready = set()
+        self.bcast = set()
+        self.scale = scale
+        self.queues = []
+        for r in range(world):
+            p = plans[r]
+            self.queues.append({
+                "red": list(p["copy"]) + list(p["reduce"]),
+                "red_kind": ["copy"] * len(p["copy"]) + ["reduce"] * len(p["reduce"]),
+                "red_i": 0,
+                "bc": list(p["bcast"]),
+                "bc_i": 0,
+            })
 
-    def buf(kind, r):
-        return {SEND: send, ACC: acc, RESULT: result}[kind][r]
+    def buf(self, kind, r):
+        return {SEND: self.send, ACC: self.acc, RESULT: self.result}[kind][r]
 
-    # per-rank sequential queues: copy -> reduce stream; bcast stream
-    queues = []
-    for r in range(world):
-        p = plans[r]
-        queues.append({
-            "red": list(p["copy"]) + list(p["reduce"]),  # same stream order
-            "red_kind": ["copy"] * len(p["copy"]) + ["reduce"] * len(p["reduce"]),
-            "red_i": 0,
-            "bc": list(p["bcast"]),
-            "bc_i": 0,
-        })
-
-    def try_red(r):
-        q = queues[r]
+    def try_red(self, r):
+        q = self.queues[r]
         i = q["red_i"]
         if i >= len(q["red"]):
             return False
         u, kind = q["red"][i], q["red_kind"][i]
         t, c = u["tree"], u["chunk"]
-        off, cnt = u["offset"], u["count"]
         if kind == "copy":
-            send[r][off:off + cnt] = user[r][off:off + cnt]
-            if u["notify"]:
-                ready.add((u["consumer"], r, t, c))
+            off, cnt = u["offset"], u["count"]
+            self.send[r][off:off + cnt] = self.user[r][off:off + cnt]
+            for peer in u["notify_to"]:
+                if u["flag_space"] == 0:
+                    self.ready.add((peer, r, t, c))
+                else:
+                    self.bcast.add((peer, t, c))
         else:
+            off, cnt = u["offset"], u["count"]
             for (sr, sk) in u["srcs"]:
-                if sr != r and (r, sr, t, c) not in ready:
+                if sr != r and (r, sr, t, c) not in self.ready:
                     return False
-            pieces = [buf(sk, sr)[off:off + cnt] for (sr, sk) in u["srcs"]]
+            pieces = [self.buf(sk, sr)[off:off + cnt] for (sr, sk) in u["srcs"]]
             if u["include_self"]:
-                pieces.append(send[r][off:off + cnt])
+                pieces.append(self.send[r][off:off + cnt])
             assert pieces, "reduce unit with no sources"
-            acc[r][off:off + cnt] = np.sum(pieces, axis=0)
+            self.acc[r][off:off + cnt] = np.sum(pieces, axis=0)
             if u["notify"]:
-                ready.add((u["consumer"], r, t, c))
+                self.ready.add((u["consumer"], r, t, c))
             if u["is_root"]:
                 for k in u["publish_to"]:
-                    bcast.add((k, t, c))
+                    self.bcast.add((k, t, c))
         q["red_i"] += 1
         return True
 
-    def try_bc(r):
-        q = queues[r]
+    def try_bc(self, r):
+        q = self.queues[r]
         i = q["bc_i"]
         if i >= len(q["bc"]):
             return False
         u = q["bc"][i]
         t, c = u["tree"], u["chunk"]
-        off, cnt = u["offset"], u["count"]
-        if (r, t, c) not in bcast:
-            return False
+        if u["parent"] >= 0 or len(u["publish_to"]) or True:
+            # self-pulls (parent == -1) still wait on the bcast flag: the
+            # producing unit pushes to self
+            if (r, t, c) not in self.bcast:
+                return False
+        soff, doff, cnt = u["src_offset"], u["dst_offset"], u["count"]
         src_rank = r if u["parent"] < 0 else u["parent"]
-        src = buf(u["parent_kind"], src_rank)[off:off + cnt]
-        out[r][off:off + cnt] = src * scale
+        src = self.buf(u["parent_kind"], src_rank)[soff:soff + cnt]
+        self.out[r][doff:doff + cnt] = src * self.scale
         if u["forward"]:
-            result[r][off:off + cnt] = src
+            self.result[r][soff:soff + cnt] = src
             for k in u["publish_to"]:
-                bcast.add((k, t, c))
+                self.bcast.add((k, t, c))
         q["bc_i"] += 1
         return True
 
-    iters = 0
-    while True:
-        progress = False
-        for r in range(world):
-            while try_red(r):
-                progress = True
-            while try_bc(r):
-                progress = True
-        iters += 1
-        done = all(
-            q["red_i"] == len(q["red"]) and q["bc_i"] == len(q["bc"])
-            for q in queues
-        )
-        if done:
-            break
-        if not progress:
-            raise AssertionError("protocol deadlock: no runnable unit")
+    def run(self):
+        while True:
+            progress = False
+            for r in range(self.world):
+                while self.try_red(r):
+                    progress = True
+                while self.try_bc(r):
+                    progress = True
+            if all(q["red_i"] == len(q["red"]) and q["bc_i"] == len(q["bc"])
+                   for q in self.queues):
+                return
+            if not progress:
+                raise AssertionError("protocol deadlock: no runnable unit")
 
-    expect = np.sum([user[r] for r in active], axis=0) * scale
+
+def rand_inputs(world, n, seed=0):
+    rng = np.random.default_rng(seed)
+    return [rng.standard_normal(n).astype(np.float32) for _ in range(world)]
+
+
+# ---------------------------------------------------------------------------
+# allreduce
+# ---------------------------------------------------------------------------
+
+
+def sim_allreduce(parents, total, active=None, chunk_bytes=256, average=False):
+    world = len(parents[0])
+    act = active if active is not None else list(range(world))
+    plans = [core.compute_plan(parents, r, total, 4, chunk_bytes, act)
+             for r in range(world)]
+    user = rand_inputs(world, total)
+    scale = 1.0 / len(act) if average else 1.0
+    sim = Sim(world, plans, user, total, scale)
+    sim.run()
+    expect = np.sum([user[r] for r in act], axis=0) * scale
     for r in range(world):
-        np.testing.assert_allclose(out[r], expect, rtol=1e-5, atol=1e-5)
-    return out, iters
+        np.testing.assert_allclose(sim.out[r], expect, rtol=1e-5, atol=1e-5)
 
 
 @pytest.mark.parametrize("world", [2, 3, 4, 8])
 def test_stars_allreduce(world):
-    simulate(stars(world), total=1000)
+    sim_allreduce(stars(world), total=1000)
 
 
 @pytest.mark.parametrize("world", [2, 4, 8])
 def test_chains_allreduce(world):
-    simulate(chains(world, ntrees=2), total=777)
+    sim_allreduce(chains(world, ntrees=2), total=777)
 
 
 @pytest.mark.parametrize("world", [4, 8])
 def test_binary_trees_allreduce(world):
-    simulate(binary_trees(world, ntrees=3), total=513)
+    sim_allreduce(binary_trees(world, ntrees=3), total=513)
 
 
 def test_single_tree():
-    simulate([[1, -1, 1, 1]], total=300)
+    sim_allreduce([[1, -1, 1, 1]], total=300)
 
 
 def test_average():
-    simulate(stars(4), total=256, average=True)
+    sim_allreduce(stars(4), total=256, average=True)
 
 
 def test_tiny_tensor_smaller_than_slices():
-    # 10 elements, 8 trees: most trees get empty slices
-    simulate(stars(8), total=10)
+    sim_allreduce(stars(8), total=10)
 
 
 @pytest.mark.parametrize("inactive", [[0], [3], [1, 2]])
 def test_relay_star_inactive(inactive):
     world = 4
     active = [r for r in range(world) if r not in inactive]
-    simulate(stars(world), total=512, active=active)
+    sim_allreduce(stars(world), total=512, active=active)
 
 
 def test_relay_chain_passthrough():
-    # chain 0<-1<-2<-3 with 1 inactive: reducer 0 must pull 2's subtree
-    # result directly (passthrough skip), totals still correct
-    world = 4
-    parents = [[-1, 0, 1, 2]]
-    simulate(parents, total=512, active=[0, 2, 3])
+    sim_allreduce([[-1, 0, 1, 2]], total=512, active=[0, 2, 3])
 
 
 def test_relay_inactive_aggregator():
-    # binary tree root 0 inactive with two subtrees: still aggregates
-    parents = [[-1, 0, 0, 1, 1, 2, 2, 3]]
-    simulate(parents, total=640, active=[1, 2, 3, 4, 5, 6, 7])
+    sim_allreduce([[-1, 0, 0, 1, 1, 2, 2, 3]], total=640,
+                  active=[1, 2, 3, 4, 5, 6, 7])
 
 
 def test_relay_only_one_active():
-    simulate(stars(4), total=128, active=[2])
+    sim_allreduce(stars(4), total=128, active=[2])
 
 
 def test_plan_deterministic_across_ranks():
     parents = stars(4)
-    # all ranks must agree on the chunk grid
     plans = [core.compute_plan(parents, r, 5000, 4, 256) for r in range(4)]
     grids = [
-        sorted((u["tree"], u["chunk"], u["offset"], u["count"]) for u in p["bcast"])
+        sorted((u["tree"], u["chunk"], u["src_offset"], u["count"])
+               for u in p["bcast"])
         for p in plans
     ]
     assert all(g == grids[0] for g in grids)
@@ -218,6 +229,95 @@ def test_plan_deterministic_across_ranks():
 
 def test_malformed_strategy_raises():
     with pytest.raises(Exception):
-        core.compute_plan([[0, 1, 2, 3]], 0, 100, 4, 256)  # cycle, no root
+        core.compute_plan([[0, 1, 2, 3]], 0, 100, 4, 256)
     with pytest.raises(Exception):
-        core.compute_plan([[-1, -1, 0, 0]], 0, 100, 4, 256)  # two roots
+        core.compute_plan([[-1, -1, 0, 0]], 0, 100, 4, 256)
+
+
+# ---------------------------------------------------------------------------
+# other primitives
+# ---------------------------------------------------------------------------
+
+
+def prim_plans(prim, world, elems, chunk_bytes=256, root=0, active=()):
+    return [
+        core.compute_primitive_plan(prim, world, r, elems, 4, chunk_bytes,
+                                    root=root, active=list(active))
+        for r in range(world)
+    ]
+
+
+@pytest.mark.parametrize("world,root", [(2, 0), (4, 2), (8, 7)])
+def test_reduce(world, root):
+    total = 900
+    plans = prim_plans("reduce", world, total, root=root)
+    user = rand_inputs(world, total)
+    sim = Sim(world, plans, user, total)
+    sim.run()
+    expect = np.sum(user, axis=0)
+    np.testing.assert_allclose(sim.out[root], expect, rtol=1e-5, atol=1e-5)
+    for r in range(world):
+        if r != root:
+            assert np.isnan(sim.out[r]).all()  # only root receives
+
+
+def test_reduce_with_relay():
+    world, root = 4, 1
+    active = [0, 2, 3]
+    plans = prim_plans("reduce", world, 500, root=root, active=active)
+    user = rand_inputs(world, 500)
+    sim = Sim(world, plans, user, 500)
+    sim.run()
+    expect = np.sum([user[r] for r in active], axis=0)
+    np.testing.assert_allclose(sim.out[root], expect, rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.parametrize("world,root", [(2, 1), (4, 0), (8, 3)])
+def test_broadcast(world, root):
+    total = 700
+    plans = prim_plans("broadcast", world, total, root=root)
+    user = rand_inputs(world, total)
+    sim = Sim(world, plans, user, total)
+    sim.run()
+    for r in range(world):
+        if r != root:
+            np.testing.assert_allclose(sim.out[r], user[root], rtol=1e-6,
+                                       atol=1e-6)
+
+
+@pytest.mark.parametrize("world", [2, 4, 8])
+def test_allgather(world):
+    L = 300
+    plans = prim_plans("allgather", world, L)
+    user = rand_inputs(world, L)
+    sim = Sim(world, plans, user, world * L)
+    sim.run()
+    expect = np.concatenate(user)
+    for r in range(world):
+        np.testing.assert_allclose(sim.out[r], expect, rtol=1e-6, atol=1e-6)
+
+
+@pytest.mark.parametrize("world", [2, 4, 8])
+def test_reducescatter(world):
+    L = 256
+    plans = prim_plans("reducescatter", world, L)
+    user = rand_inputs(world, world * L)
+    sim = Sim(world, plans, user, L)
+    sim.run()
+    total = np.sum(user, axis=0)
+    for r in range(world):
+        np.testing.assert_allclose(sim.out[r], total[r * L:(r + 1) * L],
+                                   rtol=1e-5, atol=1e-5)
+
+
+@pytest.mark.parametrize("world", [2, 4, 8])
+def test_alltoall(world):
+    L = 200
+    plans = prim_plans("alltoall", world, L)
+    user = rand_inputs(world, world * L)
+    sim = Sim(world, plans, user, world * L)
+    sim.run()
+    for r in range(world):
+        expect = np.concatenate([user[s][r * L:(r + 1) * L]
+                                 for s in range(world)])
+        np.testing.assert_allclose(sim.out[r], expect, rtol=1e-6, atol=1e-6)

@@ -49,12 +49,15 @@ class CommArgs:
     profile_freq: int = 0     # reconstruct_topology every N steps (0 = never)
     policy: str = "par-trees"
     chunk_bytes: int = 4 * 1024 * 1024
+    relay: bool = False       # straggler-adaptive active sets via coordinator
+    coordinator_port: int = 50051
 
     @classmethod
     def from_namespace(cls, ns) -> "CommArgs":
         kw = {}
         for f in ("port", "entry_point", "strategy_file", "logical_graph",
-                  "parallel_degree", "profile_freq", "policy", "chunk_bytes"):
+                  "parallel_degree", "profile_freq", "policy", "chunk_bytes",
+                  "relay", "coordinator_port"):
             if hasattr(ns, f) and getattr(ns, f) is not None:
                 kw[f] = getattr(ns, f)
         return cls(**kw)
@@ -82,7 +85,9 @@ class Communicator:
         self.use_gpu = torch.cuda.is_available()
         self.coordinator = None   # rank-0 gRPC server (relay/fault)
         self.controller = None    # per-rank controller thread
+        self.hooker = None        # per-rank hook-negotiation client
         self.active_ranks: Optional[List[int]] = None  # None = all
+        self.fault_worker_list: List[int] = []
         self._setup_done = False
 
         self.synthesizer = Synthesizer(
@@ -172,9 +177,57 @@ class Communicator:
             self.engine.set_strategy(self.strategy)
         else:
             raise ValueError(f"unknown ADAPCC_TRANSPORT={transport}")
+        if self.args.relay and self.world_size > 1:
+            self._start_relay_control()
         self._setup_done = True
         log.info("[Rank %d] transmission context setup time: %.1f ms",
                  self.rank, 1000 * (time.time() - t0))
+
+    # ------------------------------------------------------------------
+    # Relay control / fault detection (reference: commu.py:81-97, proto/)
+    # ------------------------------------------------------------------
+
+    def coordinator_address(self) -> str:
+        host = os.environ.get("MASTER_ADDR", "127.0.0.1")
+        return f"{host}:{self.args.coordinator_port}"
+
+    def _start_relay_control(self) -> None:
+        from .coordinator.client import Controller, Hooker
+        from .coordinator.server import CoordinatorServer
+
+        if self.rank == 0 and self.coordinator is None:
+            self.coordinator = CoordinatorServer(
+                self.world_size, port=self.args.coordinator_port).start()
+        if dist.is_initialized():
+            dist.barrier(group=self.group)  # server up before clients dial
+        addr = self.coordinator_address()
+        if self.hooker is None:
+            self.hooker = Hooker(addr, self.rank)
+        if self.controller is None:
+            self.controller = Controller(
+                addr, self.rank,
+                on_active=self._on_controller_active,
+                on_fault=self._on_fault,
+            )
+
+    def _on_controller_active(self, active: Optional[List[int]]) -> None:
+        pass  # the hook-side negotiation governs the active set
+
+    def _on_fault(self, dead: List[int]) -> None:
+        self.fault_worker_list = dead
+        log.error("[Rank %d] fault_worker_list=%s — restart or "
+                  "reconstruct_topology with the surviving ranks",
+                  self.rank, dead)
+
+    def notify_hook_ready(self, step: int) -> None:
+        """First DDP bucket of a step: negotiate the straggler-adaptive
+        active set (reference: commu.py:387-394 + rpc_server hook_fetch)."""
+        if self.hooker is None:
+            return
+        active = self.hooker.send_ready_request(step)
+        self.active_ranks = (
+            None if len(active) >= self.world_size else active
+        )
 
     def ensure_setup(self) -> None:
         if not self._setup_done:

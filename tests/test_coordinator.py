@@ -1,0 +1,160 @@
+"""Coordinator (relay negotiation + fault detection) tests."""
+
+import os
+import threading
+import time
+
+import torch
+
+from util_mp import run_mp
+
+from adapcc_amd.coordinator.server import CoordinatorServer, CoordinatorServicer
+
+
+def test_rent_or_buy_excludes_straggler():
+    sv = CoordinatorServicer(world_size=4, time_slot=0.002,
+                             relay_threshold=0.05, fault_timeout=2.0)
+    results = {}
+
+    def worker(rank, delay):
+        time.sleep(delay)
+        results[rank] = sv.hook_fetch({"step": 3, "rank": rank})
+
+    ts = [threading.Thread(target=worker, args=(r, 0.0)) for r in range(3)]
+    straggler = threading.Thread(target=worker, args=(3, 0.5))
+    for t in ts:
+        t.start()
+    straggler.start()
+    for t in ts + [straggler]:
+        t.join()
+    active = results[0]["active"]
+    assert sorted(active) == [0, 1, 2]
+    for r in range(3):
+        assert results[r]["active"] == active
+    # the straggler opened a NEW negotiation for the step after the snapshot
+    # was taken; it sees at least itself
+    assert 3 in results[3]["active"] or results[3]["active"] == active
+
+
+def test_all_arrive_fast_full_set():
+    sv = CoordinatorServicer(world_size=3, time_slot=0.002,
+                             relay_threshold=0.2, fault_timeout=2.0)
+    results = {}
+
+    def worker(rank):
+        results[rank] = sv.hook_fetch({"step": 0, "rank": rank})
+
+    ts = [threading.Thread(target=worker, args=(r,)) for r in range(3)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    for r in range(3):
+        assert sorted(results[r]["active"]) == [0, 1, 2]
+        assert results[r]["status"] == 1
+
+
+def test_controller_fault_timeout():
+    sv = CoordinatorServicer(world_size=2, fault_timeout=0.2)
+    resp = sv.controller_fetch({"step": 5, "rank": 0})
+    assert resp["status"] == 0
+    assert resp["active"] == [0]
+
+
+def test_controller_all_alive():
+    sv = CoordinatorServicer(world_size=2, fault_timeout=5.0)
+    out = {}
+
+    def beat(rank):
+        out[rank] = sv.controller_fetch({"step": 1, "rank": rank})
+
+    ts = [threading.Thread(target=beat, args=(r,)) for r in range(2)]
+    for t in ts:
+        t.start()
+    for t in ts:
+        t.join()
+    assert out[0]["status"] == 1 and out[1]["status"] == 1
+    assert out[0]["active"] == [0, 1]
+
+
+def test_grpc_end_to_end():
+    from adapcc_amd.coordinator.client import Controller, Hooker
+
+    server = CoordinatorServer(world_size=2, port=0, time_slot=0.002,
+                               relay_threshold=0.05, fault_timeout=3.0)
+    server.start()
+    addr = f"127.0.0.1:{server.bound_port}"
+    try:
+        h0, h1 = Hooker(addr, 0), Hooker(addr, 1)
+        res = {}
+        t0 = threading.Thread(
+            target=lambda: res.setdefault(0, h0.send_ready_request(0)))
+        t1 = threading.Thread(
+            target=lambda: res.setdefault(1, h1.send_ready_request(0)))
+        t0.start(); t1.start(); t0.join(); t1.join()
+        assert sorted(res[0]) == [0, 1]
+
+        seen = []
+        faults = []
+        c0 = Controller(addr, 0, on_active=seen.append, on_fault=faults.append)
+        c1 = Controller(addr, 1, on_active=lambda a: None,
+                        on_fault=lambda d: None)
+        c0.submit_step(1)
+        c1.submit_step(1)
+        deadline = time.time() + 5
+        while not seen and time.time() < deadline:
+            time.sleep(0.01)
+        assert seen and sorted(seen[0]) == [0, 1]
+        c0.stop(); c1.stop()
+        h0.close(); h1.close()
+    finally:
+        server.stop()
+
+
+def _relay_e2e(rank, world):
+    """Facade with relay=True over gloo: rank 1 stalls before its first
+    bucket; the active set excludes it and the allreduce still completes
+    with the stragglers' contribution dropped."""
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.coordinator import server as srv
+
+    srv.TIME_SLOT_DURATION = 0.002
+    args = CommArgs(entry_point=-1, relay=True, coordinator_port=0)
+    # port 0 -> ephemeral; share the bound port via the file system store
+    import tempfile
+
+    portfile = os.path.join(tempfile.gettempdir(), "adapcc_test_coord_port")
+    if rank == 0:
+        from adapcc_amd.coordinator.server import CoordinatorServer
+
+        coord = CoordinatorServer(world, port=0, time_slot=0.002,
+                                  relay_threshold=0.1).start()
+        with open(portfile, "w") as f:
+            f.write(str(coord.bound_port))
+    import torch.distributed as dist
+
+    dist.barrier()
+    with open(portfile) as f:
+        port = int(f.read())
+    args.coordinator_port = port
+    AdapCC.init(args, rank, rank, world)
+    comm = AdapCC.communicator
+    if rank == 0:
+        comm.coordinator = coord  # adopt instead of re-binding
+    AdapCC.setup()
+
+    if rank == 1:
+        time.sleep(0.4)  # miss the rent-or-buy window
+    comm.notify_hook_ready(step=0)
+    t = torch.full((64,), float(rank + 1))
+    AdapCC.allreduce(t, average=True)
+    if rank == 0:
+        assert comm.active_ranks == [0], comm.active_ranks
+        assert torch.allclose(t, torch.ones_like(t)), t[0]
+    AdapCC.clear()
+    return True
+
+
+def test_relay_e2e_straggler():
+    assert all(run_mp(_relay_e2e, 2, backend="gloo", timeout=120))

@@ -133,6 +133,6 @@ class GPT2(nn.Module):
         loss = None
         if targets is not None:
             loss = F.cross_entropy(
-                logits.view(-1, logits.size(-1)).float(), targets.view(-1)
+                logits.view(-1, logits.size(-1)), targets.view(-1)
             )
         return logits, loss

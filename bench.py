@@ -20,6 +20,8 @@ import os
 import sys
 import time
 
+os.environ.setdefault("GPU_MAX_HW_QUEUES", "8")  # comm/compute overlap
+
 import torch
 import torch.distributed as dist
 
@@ -76,7 +78,7 @@ def main() -> None:
 
     if world > 1:
         ddp = torch.nn.parallel.DistributedDataParallel(
-            model, device_ids=[local_rank] if use_cuda else None,
+            model, device_ids=[device.index] if use_cuda else None,
             bucket_cap_mb=args.bucket_mb,
         )
         state = AdapccDDPState(AdapCC.communicator)
@@ -86,8 +88,14 @@ def main() -> None:
         state = None
         train_mod = model
 
-    opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4, betas=(0.9, 0.95),
-                            weight_decay=0.1)
+    try:
+        opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4,
+                                betas=(0.9, 0.95), weight_decay=0.1,
+                                fused=use_cuda)
+    except (RuntimeError, ValueError):
+        opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4,
+                                betas=(0.9, 0.95), weight_decay=0.1,
+                                foreach=True)
 
     B, T = args.batch, min(args.seq, cfg.n_positions)
     data = torch.randint(0, cfg.vocab_size, (B, T + 1), device=device)

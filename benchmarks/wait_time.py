@@ -61,7 +61,7 @@ def main():
     AdapCC.init(CommArgs(entry_point=-1, relay=args.relay), local_rank, rank,
                 world)
     AdapCC.setup()
-    ddp = DDP(model, device_ids=[local_rank] if use_cuda else None,
+    ddp = DDP(model, device_ids=[device.index] if use_cuda else None,
               bucket_cap_mb=25)
     state = AdapccDDPState(AdapCC.communicator)
 

@@ -118,7 +118,7 @@ def main():
     AdapCC.init(CommArgs(entry_point=-1), local_rank, rank, world)
     AdapCC.setup()
     if world > 1:
-        ddp = DDP(model, device_ids=[local_rank] if use_cuda else None)
+        ddp = DDP(model, device_ids=[device.index] if use_cuda else None)
         hstate = AdapccDDPState(AdapCC.communicator)
         ddp.register_comm_hook(hstate, adapcc_allreduce_hook)
     else:

@@ -62,7 +62,7 @@ def main():
     AdapCC.setup()
 
     if world > 1:
-        model = DDP(model, device_ids=[local_rank] if use_cuda else None,
+        model = DDP(model, device_ids=[device.index] if use_cuda else None,
                     bucket_cap_mb=100)
         state = AdapccDDPState(AdapCC.communicator)
         model.register_comm_hook(state, adapcc_allreduce_hook)

@@ -242,12 +242,6 @@ __global__ void __launch_bounds__(256) reduce_kernel(
   FlagInbox* my_inbox = tabs.inbox[me];
   const unsigned long long deadline = realtime() + args.timeout_ticks;
 
-  if (threadIdx.x == 0 && blockIdx.x == 0) {
-    __hip_atomic_store(&my_inbox->trace[0], realtime(), __ATOMIC_RELAXED,
-                       __HIP_MEMORY_SCOPE_SYSTEM);
-    __hip_atomic_store(&my_inbox->trace[1], args.timeout_ticks, __ATOMIC_RELAXED,
-                       __HIP_MEMORY_SCOPE_SYSTEM);
-  }
   for (int ui = group; ui < n_units; ui += n_groups) {
     const ReduceUnit u = units[ui];
 
@@ -260,9 +254,6 @@ __global__ void __launch_bounds__(256) reduce_kernel(
                         my_inbox, kErrTimeoutReady,
                         ((uint64_t)u.tree << 32) | (uint32_t)u.chunk);
     }
-    if (threadIdx.x == 0)
-      __hip_atomic_store(&my_inbox->trace[2], 100 + (ok ? 1 : 0),
-                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
     if (!ok) return;
 
     // Gather source pointers.
@@ -309,13 +300,7 @@ __global__ void __launch_bounds__(256) reduce_kernel(
     }
 
     // Arrive; last workgroup publishes.
-    if (threadIdx.x == 0)
-      __hip_atomic_store(&my_inbox->trace[3], 200 + blockIdx.x,
-                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
     const bool last = unit_arrive(&counters[ui], wgs_per_group);
-    if (threadIdx.x == 0 && last)
-      __hip_atomic_store(&my_inbox->trace[4], 300 + blockIdx.x,
-                         __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_SYSTEM);
     if (last && threadIdx.x == 0) {
       if (u.notify_parent) {
         push_flag(&tabs.inbox[u.parent_rank]->ready[me][u.tree][u.chunk], args.seq);

@@ -42,6 +42,60 @@ def parse_sizes(spec: str):
 DEFAULT_SIZES = "4K,16K,64K,256K,1M,4M,16M,64M,256M,1G"
 
 
+def _op_fn_native(eng, t, op, world):
+    import torch as _t
+
+    if op == "allreduce":
+        return lambda: eng.all_reduce(t)
+    if op == "broadcast":
+        return lambda: eng.broadcast(t, root=0)
+    if op == "reduce":
+        return lambda: eng.reduce(t, root=0)
+    if op == "allgather":
+        out = _t.empty(t.numel() * world, dtype=t.dtype, device=t.device)
+        return lambda: eng.all_gather(out, t)
+    if op == "alltoall":
+        out = _t.empty_like(t)
+        return lambda: eng.all_to_all(out, t)
+    if op == "reducescatter":
+        out = _t.empty(t.numel() // world, dtype=t.dtype, device=t.device)
+        return lambda: eng.reduce_scatter(out, t)
+    raise ValueError(op)
+
+
+def _op_fn_pg(t, op, world):
+    import torch as _t
+
+    if op == "allreduce":
+        return lambda: dist.all_reduce(t)
+    if op == "broadcast":
+        return lambda: dist.broadcast(t, src=0)
+    if op == "reduce":
+        return lambda: dist.reduce(t, dst=0)
+    if op == "allgather":
+        out = _t.empty(t.numel() * world, dtype=t.dtype, device=t.device)
+        return lambda: dist.all_gather_into_tensor(out, t)
+    if op == "alltoall":
+        out = _t.empty_like(t)
+        return lambda: dist.all_to_all_single(out, t)
+    if op == "reducescatter":
+        out = _t.empty(t.numel() // world, dtype=t.dtype, device=t.device)
+        return lambda: dist.reduce_scatter_tensor(out, t)
+    raise ValueError(op)
+
+
+# busbw factors per PERFORMANCE.md: allreduce 2(n-1)/n; AG/RS/alltoall
+# (n-1)/n; broadcast/reduce 1
+def bus_factor(op, world):
+    if world <= 1:
+        return 1.0
+    if op == "allreduce":
+        return 2 * (world - 1) / world
+    if op in ("allgather", "reducescatter", "alltoall"):
+        return (world - 1) / world
+    return 1.0
+
+
 def bench_transport(make_allreduce, sizes, iters, warmup, device, world):
     rows = []
     for size in sizes:
@@ -68,8 +122,7 @@ def bench_transport(make_allreduce, sizes, iters, warmup, device, world):
             dist.all_reduce(m, op=dist.ReduceOp.MAX)
             dt = float(m)
         algbw = size / dt / 1e9
-        busbw = algbw * 2 * (world - 1) / world if world > 1 else algbw
-        rows.append((size, dt * 1e6, algbw, busbw))
+        rows.append((size, dt * 1e6, algbw))
     return rows
 
 
@@ -79,6 +132,9 @@ def main():
     p.add_argument("--iters", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--transports", default="native,pg")
+    p.add_argument("--op", default="allreduce",
+                   choices=["allreduce", "allgather", "alltoall",
+                            "reducescatter", "broadcast", "reduce"])
     p.add_argument("--chunk_bytes", type=int, default=1 << 20)
     p.add_argument("--out", default="")
     args = p.parse_args()
@@ -113,31 +169,28 @@ def main():
             strat = synthesize_stars(world, chunk_bytes=args.chunk_bytes)
             eng.set_strategy(strat)
 
-            def make(t, eng=eng):
-                def fn():
-                    eng.all_reduce(t)
-                return fn
+            def make(t, eng=eng, op=args.op, world=world):
+                return _op_fn_native(eng, t, op, world)
         elif transport == "pg":
             if world == 1:
                 continue
 
-            def make(t):
-                def fn():
-                    dist.all_reduce(t)
-                return fn
+            def make(t, op=args.op, world=world):
+                return _op_fn_pg(t, op, world)
         else:
             raise SystemExit(f"unknown transport {transport}")
 
         rows = bench_transport(make, sizes, args.iters, args.warmup, device,
                                world)
-        results[transport] = rows
+        bf = bus_factor(args.op, world)
+        results[transport] = [(sz, us, ab, ab * bf) for sz, us, ab in rows]
         if transport == "native":
             eng.synchronize()
 
     if rank == 0:
         lines = ["transport,bytes,us,algbw_GBps,busbw_GBps"]
         for tr, rows in results.items():
-            print(f"\n== {tr} (n={world}) ==")
+            print(f"\n== {args.op} {tr} (n={world}) ==")
             print(f"{'bytes':>12} {'time(us)':>12} {'algbw GB/s':>12} "
                   f"{'busbw GB/s':>12}")
             for size, us, algbw, busbw in rows:
@@ -148,7 +201,7 @@ def main():
             with open(args.out, "w") as f:
                 f.write("\n".join(lines) + "\n")
         summary = {
-            "metric": "allreduce_busbw_GBps",
+            "metric": f"{args.op}_busbw_GBps",
             "n_gpus": world,
             "results": {tr: [[s, round(b, 2)] for s, _, _, b in rows]
                         for tr, rows in results.items()},

@@ -208,3 +208,63 @@ def _ddp_hook_gpu(rank, world):
 
 def test_ddp_hook_native_gpu():
     assert all(run_mp(_ddp_hook_gpu, 2, backend="gloo", timeout=300))
+
+
+def _timeout_raises(rank, world):
+    """A wedged peer must surface as a Python exception within the
+    deadline, never a hung GPU: rank 1 simply never joins the collective."""
+    os.environ["ADAPCC_TIMEOUT_MS"] = "2500"
+    import torch
+    import torch.distributed as dist
+
+    torch.cuda.set_device(0)
+    from adapcc_amd.runtime.engine import NativeEngine
+    from adapcc_amd.strategy.partrees import synthesize_stars
+
+    eng = NativeEngine(rank, world, device=0, cap_bytes=16 << 20)
+    eng.bootstrap()
+    eng.set_strategy(synthesize_stars(world))
+    raised = False
+    if rank == 0:
+        t = torch.ones(4096, device="cuda")
+        eng.all_reduce(t)
+        try:
+            eng.synchronize()
+        except RuntimeError as e:
+            raised = "timeout" in str(e) or "error" in str(e)
+    # rank 1 stays alive (its memory must remain mapped) until rank 0 is
+    # done; then both exit
+    dist.barrier()
+    return raised if rank == 0 else True
+
+
+def test_timeout_surfaces_as_exception():
+    res = run_mp(_timeout_raises, 2, backend="gloo", timeout=180)
+    assert res[0] is True
+
+
+def _strategy_switch(rank, world):
+    """reconstruct_topology path: re-set strategy on a live engine and keep
+    reducing correctly (plan cache invalidation + epoch continuity)."""
+    os.environ["ADAPCC_TIMEOUT_MS"] = "20000"
+    import torch
+
+    torch.cuda.set_device(0)
+    from adapcc_amd.runtime.engine import NativeEngine
+    from adapcc_amd.strategy.partrees import synthesize_chains, synthesize_stars
+
+    eng = NativeEngine(rank, world, device=0, cap_bytes=16 << 20)
+    eng.bootstrap()
+    for strat in (synthesize_stars(world), synthesize_chains(world, 2),
+                  synthesize_stars(world)):
+        eng.set_strategy(strat)
+        t = torch.full((10_000,), float(rank + 1), device="cuda")
+        eng.all_reduce(t)
+        eng.synchronize()
+        expect = float(sum(range(1, world + 1)))
+        assert torch.allclose(t, torch.full_like(t, expect)), t[:3]
+    return True
+
+
+def test_strategy_switch_live():
+    assert all(run_mp(_strategy_switch, 2, backend="gloo", timeout=180))

@@ -88,6 +88,7 @@ class Communicator:
         self.hooker = None        # per-rank hook-negotiation client
         self.active_ranks: Optional[List[int]] = None  # None = all
         self.fault_worker_list: List[int] = []
+        self.effective_transport: Optional[str] = None
         self._setup_done = False
 
         self.synthesizer = Synthesizer(
@@ -157,24 +158,25 @@ class Communicator:
             self.run_entry_point()
         t0 = time.time()
         transport = self.transport
+        multi_node = self.graph is not None and len(self.graph.servers) > 1
         if transport == "auto":
-            transport = "native" if self.use_gpu else "pg"
+            transport = "native" if (self.use_gpu and not multi_node) else "pg"
+        if transport == "native" and multi_node:
+            log.warning("native engine is single-node; multi-node runs use "
+                        "the process-group transport")
 
         if transport == "native":
-            from .runtime.engine import NativeEngine
-
-            torch.cuda.set_device(self.local_rank)
-            self.engine = NativeEngine(self.rank, self.world_size,
-                                       device=self.local_rank)
-            self.engine.bootstrap(group=self.group)
-            self.engine.set_strategy(self.strategy)
-            self.engine.self_test()
+            try:
+                self._setup_native()
+            except Exception as e:
+                if self.transport != "auto":
+                    raise
+                log.error("NATIVE ENGINE SETUP FAILED (%s); falling back to "
+                          "the process-group transport — performance and "
+                          "the custom data plane are NOT in effect", e)
+                self._setup_pg()
         elif transport == "pg":
-            from .runtime.fallback import ProcessGroupEngine
-
-            self.engine = ProcessGroupEngine(self.rank, self.world_size,
-                                             group=self.group)
-            self.engine.set_strategy(self.strategy)
+            self._setup_pg()
         else:
             raise ValueError(f"unknown ADAPCC_TRANSPORT={transport}")
         if self.args.relay and self.world_size > 1:
@@ -228,6 +230,25 @@ class Communicator:
         self.active_ranks = (
             None if len(active) >= self.world_size else active
         )
+
+    def _setup_native(self) -> None:
+        from .runtime.engine import NativeEngine
+
+        torch.cuda.set_device(self.local_rank)
+        self.engine = NativeEngine(self.rank, self.world_size,
+                                   device=self.local_rank)
+        self.engine.bootstrap(group=self.group)
+        self.engine.set_strategy(self.strategy)
+        self.engine.self_test()
+        self.effective_transport = "native"
+
+    def _setup_pg(self) -> None:
+        from .runtime.fallback import ProcessGroupEngine
+
+        self.engine = ProcessGroupEngine(self.rank, self.world_size,
+                                         group=self.group)
+        self.engine.set_strategy(self.strategy)
+        self.effective_transport = "pg"
 
     def ensure_setup(self) -> None:
         if not self._setup_done:

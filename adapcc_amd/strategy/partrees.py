@@ -43,6 +43,8 @@ def synthesize_stars(
     chunk_bytes: int = DEFAULT_CHUNK_BYTES,
 ) -> Strategy:
     """Parallel star forest for a fully connected intra-node mesh."""
+    if ips is not None and not isinstance(ips, (list, tuple)):
+        raise TypeError("ips must be a list of addresses")
     if ips is None:
         ips = ["127.0.0.1"] * world_size
     if num_trees is None:
@@ -66,6 +68,8 @@ def synthesize_chains(
 ) -> Strategy:
     """Parallel pipelined chains (rotated). Mostly useful as a baseline and
     for bandwidth-asymmetric meshes; the star forest dominates on xGMI."""
+    if ips is not None and not isinstance(ips, (list, tuple)):
+        raise TypeError("ips must be a list of addresses")
     if ips is None:
         ips = ["127.0.0.1"] * world_size
     trees = []

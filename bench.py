@@ -161,7 +161,7 @@ def main() -> None:
                 "seq_len": T,
                 "parallelism": f"dp{world}",
                 "bucket_cap_mb": args.bucket_mb,
-                "transport": os.environ.get("ADAPCC_TRANSPORT", "auto"),
+                "transport": AdapCC.communicator.effective_transport or os.environ.get("ADAPCC_TRANSPORT", "auto"),
             },
         }
         print(json.dumps(out), flush=True)

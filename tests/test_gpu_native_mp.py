@@ -255,7 +255,8 @@ def _strategy_switch(rank, world):
 
     eng = NativeEngine(rank, world, device=0, cap_bytes=16 << 20)
     eng.bootstrap()
-    for strat in (synthesize_stars(world), synthesize_chains(world, 2),
+    for strat in (synthesize_stars(world),
+                  synthesize_chains(world, num_trees=2),
                   synthesize_stars(world)):
         eng.set_strategy(strat)
         t = torch.full((10_000,), float(rank + 1), device="cuda")

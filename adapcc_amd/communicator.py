@@ -48,7 +48,7 @@ class CommArgs:
     parallel_degree: int = 0  # 0 = auto (world_size stars on one node)
     profile_freq: int = 0     # reconstruct_topology every N steps (0 = never)
     policy: str = "par-trees"
-    chunk_bytes: int = 4 * 1024 * 1024
+    chunk_bytes: int = 2 * 1024 * 1024
     relay: bool = False       # straggler-adaptive active sets via coordinator
     coordinator_port: int = 50051
 

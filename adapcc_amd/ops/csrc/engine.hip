@@ -99,17 +99,17 @@ Engine::~Engine() {
       if (r != rank_ && peer_base_[r]) (void)hipIpcCloseMemHandle(peer_base_[r]);
     }
   }
-  if (region_) hipFree(region_);
-  if (counters_) hipFree(counters_);
-  if (h_err_) hipHostFree(h_err_);
+  if (region_) (void)hipFree(region_);
+  if (counters_) (void)hipFree(counters_);
+  if (h_err_) (void)hipHostFree(h_err_);
   for (auto& kv : plans_) kv.second.free_device();
-  hipStreamDestroy(s_red_);
-  hipStreamDestroy(s_bcast_);
-  hipStreamDestroy(s_err_);
-  hipEventDestroy(ev_in_);
-  hipEventDestroy(ev_sync0_);
-  hipEventDestroy(ev_red_);
-  hipEventDestroy(ev_barrier_);
+  (void)hipStreamDestroy(s_red_);
+  (void)hipStreamDestroy(s_bcast_);
+  (void)hipStreamDestroy(s_err_);
+  (void)hipEventDestroy(ev_in_);
+  (void)hipEventDestroy(ev_sync0_);
+  (void)hipEventDestroy(ev_red_);
+  (void)hipEventDestroy(ev_barrier_);
 }
 
 std::string Engine::ipc_handle() const {
@@ -172,10 +172,10 @@ void Engine::set_strategy(const std::vector<std::vector<int>>& parents,
 }
 
 void Engine::Plan::free_device() {
-  if (d_c) hipFree(d_c);
-  if (d_r) hipFree(d_r);
-  if (d_b) hipFree(d_b);
-  if (d_ranks) hipFree(d_ranks);
+  if (d_c) (void)hipFree(d_c);
+  if (d_r) (void)hipFree(d_r);
+  if (d_b) (void)hipFree(d_b);
+  if (d_ranks) (void)hipFree(d_ranks);
   d_c = nullptr; d_r = nullptr; d_b = nullptr; d_ranks = nullptr;
 }
 

@@ -94,6 +94,11 @@ class AdapCC:
             cls.communicator = None
 
     @classmethod
+    def stats(cls) -> dict:
+        cls._require_init()
+        return cls.communicator.stats()
+
+    @classmethod
     def _require_init(cls) -> None:
         if cls.communicator is None:
             raise RuntimeError("AdapCC.init() has not been called")

@@ -264,7 +264,18 @@ class Communicator:
         self.ensure_setup()
         if active is None:
             active = self.active_ranks
+        from .utils.metrics import GLOBAL as metrics
+
+        metrics.inc("allreduce_calls")
+        metrics.inc("allreduce_bytes", tensor.numel() * tensor.element_size())
+        if active is not None:
+            metrics.inc("allreduce_relay_calls")
         return self.engine.all_reduce(tensor, active=active, average=average)
+
+    def stats(self) -> dict:
+        from .utils.metrics import GLOBAL as metrics
+
+        return metrics.snapshot()
 
     def synchronize(self) -> None:
         if self.engine is not None:

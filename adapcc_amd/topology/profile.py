@@ -33,10 +33,11 @@ def _device() -> torch.device:
 
 def _timed_transfer(send_to: Optional[int], recv_from: Optional[int],
                     elems: int, device, group) -> float:
-    """One timed ping: returns seconds for a one-way transfer of `elems`
-    fp32 (measured at the receiver side as half of a round trip would be
-    noisy; we time send+matching recv wall clock under a barrier)."""
+    """Timed one-way transfers with a closing ack so sender-side buffering
+    cannot under-count: the sender's clock stops only after the receiver
+    confirms the last payload."""
     t = torch.ones(elems, dtype=torch.float32, device=device)
+    ack = torch.zeros(1, dtype=torch.float32, device=device)
     if device.type == "cuda":
         torch.cuda.synchronize()
     start = time.perf_counter()
@@ -45,6 +46,10 @@ def _timed_transfer(send_to: Optional[int], recv_from: Optional[int],
             dist.send(t, dst=send_to, group=group)
         if recv_from is not None:
             dist.recv(t, src=recv_from, group=group)
+    if send_to is not None:
+        dist.recv(ack, src=send_to, group=group)
+    if recv_from is not None:
+        dist.send(ack, dst=recv_from, group=group)
     if device.type == "cuda":
         torch.cuda.synchronize()
     return (time.perf_counter() - start) / _REPS

@@ -329,3 +329,35 @@ def _moe_alltoall_gpu(rank, world):
 
 def test_moe_alltoall_native_gpu():
     assert all(run_mp(_moe_alltoall_gpu, 2, backend="gloo", timeout=180))
+
+
+def _oversized_tensor_split(rank, world):
+    """Tensors above engine capacity are transparently split into
+    capacity-sized sub-calls by the wrapper."""
+    os.environ["ADAPCC_TIMEOUT_MS"] = "20000"
+    import torch
+
+    torch.cuda.set_device(0)
+    from adapcc_amd.runtime.engine import NativeEngine
+    from adapcc_amd.strategy.partrees import synthesize_stars
+
+    eng = NativeEngine(rank, world, device=0, cap_bytes=4 << 20)
+    eng.bootstrap()
+    eng.set_strategy(synthesize_stars(world))
+    n = (10 << 20) // 4  # 10 MB >> 4 MB cap
+    torch.manual_seed(rank)
+    t = torch.randn(n, device="cuda")
+    cpu = t.cpu()
+    import torch.distributed as dist
+
+    g = [torch.zeros_like(cpu) for _ in range(world)]
+    dist.all_gather(g, cpu)
+    eng.all_reduce(t)
+    eng.synchronize()
+    torch.testing.assert_close(t.cpu(), torch.stack(g).sum(0), rtol=1e-4,
+                               atol=1e-4)
+    return True
+
+
+def test_oversized_tensor_split():
+    assert all(run_mp(_oversized_tensor_split, 2, backend="gloo", timeout=180))

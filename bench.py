@@ -45,6 +45,10 @@ def parse_args():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--transport", type=str, default=None,
                    help="override ADAPCC_TRANSPORT")
+    p.add_argument("--precision", type=str, default="bf16",
+                   choices=["autocast", "bf16"],
+                   help="autocast: fp32 params + bf16 autocast; bf16: bf16 "
+                        "params/grads with fp32 master AdamW (Megatron-style)")
     p.add_argument("--graphs", type=str, default="auto",
                    choices=["auto", "on", "off"],
                    help="capture the whole step in a hipGraph (world==1)")
@@ -74,6 +78,9 @@ def main() -> None:
     cfg = getattr(GPT2Config, args.model)()
     model = GPT2(cfg).to(device)
     n_params = model.num_params()
+    bf16_master = args.precision == "bf16" and use_cuda
+    if bf16_master:
+        model = model.to(torch.bfloat16)
 
     AdapCC.init(CommArgs(entry_point=-1, policy="par-trees"),
                 local_rank, rank, world)
@@ -93,15 +100,26 @@ def main() -> None:
 
     want_graphs = (args.graphs == "on" or
                    (args.graphs == "auto" and world == 1)) and use_cuda
-    try:
-        opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4,
-                                betas=(0.9, 0.95), weight_decay=0.1,
-                                fused=use_cuda, capturable=want_graphs)
-    except (RuntimeError, ValueError):
-        opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4,
-                                betas=(0.9, 0.95), weight_decay=0.1,
-                                foreach=True)
-        want_graphs = False
+    if bf16_master:
+        # fp32 master copies; bf16 params/grads on the model (and on the
+        # wire: the DDP hook then allreduces bf16 buckets — half traffic)
+        model_params = [p for p in train_mod.parameters() if p.requires_grad]
+        masters = [p.detach().float().clone() for p in model_params]
+        for m in masters:
+            m.grad = torch.zeros_like(m)
+        opt = torch.optim.AdamW(masters, lr=3e-4, betas=(0.9, 0.95),
+                                weight_decay=0.1, fused=True,
+                                capturable=want_graphs)
+    else:
+        try:
+            opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4,
+                                    betas=(0.9, 0.95), weight_decay=0.1,
+                                    fused=use_cuda, capturable=want_graphs)
+        except (RuntimeError, ValueError):
+            opt = torch.optim.AdamW(train_mod.parameters(), lr=3e-4,
+                                    betas=(0.9, 0.95), weight_decay=0.1,
+                                    foreach=True)
+            want_graphs = False
 
     B, T = args.batch, min(args.seq, cfg.n_positions)
     data = torch.randint(0, cfg.vocab_size, (B, T + 1), device=device)
@@ -114,11 +132,22 @@ def main() -> None:
     def eager_step(i: int) -> None:
         if state is not None:
             state.on_step(i)
-        opt.zero_grad(set_to_none=True)
-        with autocast:
+        if bf16_master:
+            for p in model_params:
+                p.grad = None
             _, loss = train_mod(x, y)
-        loss.backward()
-        opt.step()
+            loss.backward()
+            torch._foreach_copy_([m.grad for m in masters],
+                                 [p.grad for p in model_params])
+            opt.step()
+            with torch.no_grad():
+                torch._foreach_copy_(model_params, masters)
+        else:
+            opt.zero_grad(set_to_none=True)
+            with autocast:
+                _, loss = train_mod(x, y)
+            loss.backward()
+            opt.step()
 
     step = eager_step
     use_graphs = want_graphs
@@ -133,15 +162,28 @@ def main() -> None:
                 with torch.cuda.stream(s):
                     eager_step(i)
                 torch.cuda.current_stream().wait_stream(s)
-            opt.zero_grad(set_to_none=False)  # grads stay allocated
             torch.cuda.synchronize()
             g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
-                with autocast:
+            if bf16_master:
+                with torch.cuda.graph(g):
+                    for p in model_params:
+                        if p.grad is not None:
+                            p.grad.zero_()
                     _, loss = train_mod(x, y)
-                loss.backward()
-                opt.step()
-                opt.zero_grad(set_to_none=False)
+                    loss.backward()
+                    torch._foreach_copy_([m.grad for m in masters],
+                                         [p.grad for p in model_params])
+                    opt.step()
+                    with torch.no_grad():
+                        torch._foreach_copy_(model_params, masters)
+            else:
+                opt.zero_grad(set_to_none=False)  # grads stay allocated
+                with torch.cuda.graph(g):
+                    with autocast:
+                        _, loss = train_mod(x, y)
+                    loss.backward()
+                    opt.step()
+                    opt.zero_grad(set_to_none=False)
 
             def graph_step(i: int) -> None:
                 g.replay()
@@ -201,6 +243,7 @@ def main() -> None:
                 "parallelism": f"dp{world}",
                 "bucket_cap_mb": args.bucket_mb,
                 "hipgraph": bool(use_graphs),
+                "precision": args.precision,
                 "transport": AdapCC.communicator.effective_transport or os.environ.get("ADAPCC_TRANSPORT", "auto"),
             },
         }

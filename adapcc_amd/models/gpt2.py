@@ -17,6 +17,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fused import FusedLayerNorm
+
 
 @dataclass
 class GPT2Config:
@@ -79,9 +81,9 @@ class MLP(nn.Module):
 class Block(nn.Module):
     def __init__(self, cfg: GPT2Config):
         super().__init__()
-        self.ln_1 = nn.LayerNorm(cfg.n_embd)
+        self.ln_1 = FusedLayerNorm(cfg.n_embd)
         self.attn = CausalSelfAttention(cfg)
-        self.ln_2 = nn.LayerNorm(cfg.n_embd)
+        self.ln_2 = FusedLayerNorm(cfg.n_embd)
         self.mlp = MLP(cfg)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -98,7 +100,7 @@ class GPT2(nn.Module):
         self.wpe = nn.Embedding(cfg.n_positions, cfg.n_embd)
         self.drop = nn.Dropout(cfg.dropout)
         self.h = nn.ModuleList(Block(cfg) for _ in range(cfg.n_layer))
-        self.ln_f = nn.LayerNorm(cfg.n_embd)
+        self.ln_f = FusedLayerNorm(cfg.n_embd)
         self.lm_head = nn.Linear(cfg.n_embd, cfg.vocab_size, bias=False)
         self.lm_head.weight = self.wte.weight  # weight tying (as GPT-2)
         self.apply(self._init)

@@ -9,6 +9,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fused import FusedLayerNorm
+
 
 @dataclass
 class ViTConfig:
@@ -33,11 +35,11 @@ class ViTConfig:
 class EncoderBlock(nn.Module):
     def __init__(self, dim: int, heads: int, mlp_dim: int):
         super().__init__()
-        self.ln1 = nn.LayerNorm(dim)
+        self.ln1 = FusedLayerNorm(dim)
         self.qkv = nn.Linear(dim, 3 * dim)
         self.proj = nn.Linear(dim, dim)
         self.heads = heads
-        self.ln2 = nn.LayerNorm(dim)
+        self.ln2 = FusedLayerNorm(dim)
         self.mlp = nn.Sequential(
             nn.Linear(dim, mlp_dim), nn.GELU(), nn.Linear(mlp_dim, dim)
         )
@@ -70,7 +72,7 @@ class ViT(nn.Module):
             EncoderBlock(cfg.dim, cfg.heads, cfg.mlp_dim)
             for _ in range(cfg.depth)
         )
-        self.ln = nn.LayerNorm(cfg.dim)
+        self.ln = FusedLayerNorm(cfg.dim)
         self.head = nn.Linear(cfg.dim, cfg.num_classes)
         nn.init.trunc_normal_(self.pos, std=0.02)
         nn.init.trunc_normal_(self.cls_token, std=0.02)

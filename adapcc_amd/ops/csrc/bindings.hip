@@ -72,6 +72,14 @@ namespace adapcc {
 void launch_local_reduce(Dtype dt, void* dst, const void* const* srcs_dev,
                          int nsrc, long count, RedOp op, float scale,
                          hipStream_t stream);
+bool ln_supported_api(long cols, int dtype);
+void ln_forward(int dtype, const void* x, const void* w, const void* b,
+                void* y, float* mean, float* rstd, long rows, long cols,
+                float eps, hipStream_t stream);
+void ln_backward(int dtype, const void* dy, const void* x, const void* w,
+                 const float* mean, const float* rstd, void* dx,
+                 float* ws_gamma, float* ws_beta, void* dgamma, void* dbeta,
+                 long rows, long cols, int nblocks, hipStream_t stream);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -101,6 +109,34 @@ PYBIND11_MODULE(_core, m) {
         },
         py::arg("dst"), py::arg("srcs"), py::arg("count"), py::arg("dtype"),
         py::arg("op"), py::arg("scale") = 1.0f, py::arg("stream") = 0);
+
+  m.def("ln_supported", &adapcc::ln_supported_api, py::arg("cols"),
+        py::arg("dtype"));
+  m.def("ln_fwd",
+        [](int dtype, uintptr_t x, uintptr_t w, uintptr_t b, uintptr_t y,
+           uintptr_t mean, uintptr_t rstd, long rows, long cols, float eps,
+           uintptr_t stream) {
+          adapcc::ln_forward(dtype, (const void*)x, (const void*)w,
+                             (const void*)b, (void*)y, (float*)mean,
+                             (float*)rstd, rows, cols, eps,
+                             reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+  m.def("ln_bwd",
+        [](int dtype, uintptr_t dy, uintptr_t x, uintptr_t w, uintptr_t mean,
+           uintptr_t rstd, uintptr_t dx, uintptr_t wsg, uintptr_t wsb,
+           uintptr_t dgamma, uintptr_t dbeta, long rows, long cols,
+           int nblocks, uintptr_t stream) {
+          adapcc::ln_backward(dtype, (const void*)dy, (const void*)x,
+                              (const void*)w, (const float*)mean,
+                              (const float*)rstd, (void*)dx, (float*)wsg,
+                              (float*)wsb, (void*)dgamma, (void*)dbeta, rows,
+                              cols, nblocks,
+                              reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
 
   m.def("compute_plan",
         [](const std::vector<std::vector<int>>& parents, int rank,

@@ -1,0 +1,96 @@
+"""Fused CDNA4 op wrappers (autograd integration for the hand-written HIP
+kernels in csrc/lnorm.hip)."""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+_CORE = None
+_DT = {}
+
+
+def _core():
+    global _CORE
+    if _CORE is None:
+        try:
+            from adapcc_amd import _core as c
+
+            _CORE = c
+            _DT.update({torch.float32: c.DTYPE_F32, torch.float16: c.DTYPE_F16,
+                        torch.bfloat16: c.DTYPE_BF16})
+        except ImportError:
+            _CORE = False
+    return _CORE
+
+
+def ln_fusable(cols: int, dtype: torch.dtype) -> bool:
+    c = _core()
+    if not c or dtype not in _DT:
+        return False
+    return bool(c.ln_supported(cols, _DT[dtype]))
+
+
+class _FusedLayerNormFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, weight: torch.Tensor,
+                bias: torch.Tensor, eps: float) -> torch.Tensor:
+        c = _core()
+        xc = x.contiguous()
+        cols = xc.shape[-1]
+        rows = xc.numel() // cols
+        y = torch.empty_like(xc)
+        mean = torch.empty(rows, dtype=torch.float32, device=xc.device)
+        rstd = torch.empty_like(mean)
+        stream = torch.cuda.current_stream(xc.device).cuda_stream
+        c.ln_fwd(_DT[xc.dtype], xc.data_ptr(), weight.data_ptr(),
+                 bias.data_ptr(), y.data_ptr(), mean.data_ptr(),
+                 rstd.data_ptr(), rows, cols, eps, stream)
+        ctx.save_for_backward(xc, weight, mean, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        c = _core()
+        x, weight, mean, rstd = ctx.saved_tensors
+        cols = x.shape[-1]
+        rows = x.numel() // cols
+        dyc = dy.contiguous()
+        dx = torch.empty_like(x)
+        dgamma = torch.empty_like(weight)
+        dbeta = torch.empty_like(weight)
+        nblocks = max(1, min(1024, (rows + 3) // 4))
+        ws = torch.empty(2 * nblocks * cols, dtype=torch.float32,
+                         device=x.device)
+        wsg = ws[: nblocks * cols]
+        wsb = ws[nblocks * cols:]
+        stream = torch.cuda.current_stream(x.device).cuda_stream
+        c.ln_bwd(_DT[x.dtype], dyc.data_ptr(), x.data_ptr(),
+                 weight.data_ptr(), mean.data_ptr(), rstd.data_ptr(),
+                 dx.data_ptr(), wsg.data_ptr(), wsb.data_ptr(),
+                 dgamma.data_ptr(), dbeta.data_ptr(), rows, cols, nblocks,
+                 stream)
+        return dx, dgamma, dbeta, None
+
+
+class FusedLayerNorm(nn.LayerNorm):
+    """Drop-in nn.LayerNorm that runs the hand-written CDNA4 kernels when
+    the shape/dtype qualify (GPU, matching weight dtype, supported width);
+    falls back to the stock implementation otherwise (CPU, autocast-mixed
+    dtypes, odd widths)."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if (
+            x.is_cuda
+            and not torch.is_autocast_enabled()
+            and len(self.normalized_shape) == 1
+            and self.weight is not None
+            and self.bias is not None
+            and self.weight.dtype == x.dtype
+            and ln_fusable(x.shape[-1], x.dtype)
+        ):
+            return _FusedLayerNormFn.apply(x, self.weight, self.bias,
+                                           self.eps)
+        return super().forward(x)

@@ -78,8 +78,8 @@ void ln_forward(int dtype, const void* x, const void* w, const void* b,
                 float eps, hipStream_t stream);
 void ln_backward(int dtype, const void* dy, const void* x, const void* w,
                  const float* mean, const float* rstd, void* dx,
-                 float* ws_gamma, float* ws_beta, void* dgamma, void* dbeta,
-                 long rows, long cols, int nblocks, hipStream_t stream);
+                 float* ws_gamma, float* ws_beta, long rows, long cols,
+                 int nblocks, hipStream_t stream);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -126,13 +126,11 @@ PYBIND11_MODULE(_core, m) {
   m.def("ln_bwd",
         [](int dtype, uintptr_t dy, uintptr_t x, uintptr_t w, uintptr_t mean,
            uintptr_t rstd, uintptr_t dx, uintptr_t wsg, uintptr_t wsb,
-           uintptr_t dgamma, uintptr_t dbeta, long rows, long cols,
-           int nblocks, uintptr_t stream) {
+           long rows, long cols, int nblocks, uintptr_t stream) {
           adapcc::ln_backward(dtype, (const void*)dy, (const void*)x,
                               (const void*)w, (const float*)mean,
                               (const float*)rstd, (void*)dx, (float*)wsg,
-                              (float*)wsb, (void*)dgamma, (void*)dbeta, rows,
-                              cols, nblocks,
+                              (float*)wsb, rows, cols, nblocks,
                               reinterpret_cast<hipStream_t>(stream));
           hipError_t e = hipGetLastError();
           if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));

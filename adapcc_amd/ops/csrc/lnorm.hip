@@ -6,8 +6,9 @@
 // registers between the statistics and normalize passes (COLS is a
 // template parameter so all register indexing is static), cross-lane sums
 // use shfl_xor, and the backward fuses dx with per-block dgamma/dbeta
-// partials (single read of dy/x) flushed through an LDS fp32 image; a tiny
-// second kernel folds the partials.
+// partials (single read of dy/x) accumulated in registers — each lane owns
+// a fixed column set — and flushed once per wave; a tiny second kernel
+// folds the per-wave partials.
 //
 // bf16/f16/f32 activations, fp32 statistics; instantiated for the common
 // transformer widths (128..4096, divisible by the 16-byte vector width).
@@ -179,15 +180,27 @@ __global__ void __launch_bounds__(256) ln_bwd_kernel(
   using IO = typename RS::IO;
   using Vec = typename IO::Vec;
   constexpr int PV = RS::PV;
-  __shared__ __attribute__((aligned(16))) float s_acc[2 * COLS];
-  for (int i = threadIdx.x; i < 2 * COLS; i += blockDim.x) s_acc[i] = 0.f;
-  __syncthreads();
-  float* s_gamma = s_acc;
-  float* s_beta = s_acc + COLS;
-
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const long wave_stride = (long)gridDim.x * 4;
+
+  // each lane owns the same column set for every row it touches ->
+  // dgamma/dbeta partials accumulate in registers, flushed once per wave
+  float dg[RS::kElems];
+  float db[RS::kElems];
+#pragma unroll
+  for (int i = 0; i < RS::kElems; ++i) dg[i] = db[i] = 0.f;
+
+  // weight is row-invariant: load it once
+  float wv[RS::kElems];
+#pragma unroll
+  for (int s = 0; s < RS::kChunks; ++s) {
+    const int vi = s * 64 + lane;
+    if (vi < RS::kVecs) {
+      Vec v = reinterpret_cast<const Vec*>(w)[vi];
+      IO::unpack(v, &wv[s * PV]);
+    }
+  }
 
   for (long row = (long)blockIdx.x * 4 + wave; row < rows; row += wave_stride) {
     const T* dyr = dy + row * COLS;
@@ -195,26 +208,23 @@ __global__ void __launch_bounds__(256) ln_bwd_kernel(
     const float mean = mean_in[row];
     const float rstd = rstd_in[row];
 
-    float dyv[RS::kElems], xh[RS::kElems], wv[RS::kElems];
+    float dyv[RS::kElems], xh[RS::kElems];
     float s1 = 0.f, s2 = 0.f;
 #pragma unroll
     for (int s = 0; s < RS::kChunks; ++s) {
       const int vi = s * 64 + lane;
       if (vi < RS::kVecs) {
-        float d[PV], xx[PV], ww[PV];
+        float d[PV], xx[PV];
         Vec vd = reinterpret_cast<const Vec*>(dyr)[vi];
         Vec vx = reinterpret_cast<const Vec*>(xr)[vi];
-        Vec vw = reinterpret_cast<const Vec*>(w)[vi];
         IO::unpack(vd, d);
         IO::unpack(vx, xx);
-        IO::unpack(vw, ww);
 #pragma unroll
         for (int k = 0; k < PV; ++k) {
           const float xhat = (xx[k] - mean) * rstd;
-          const float a = d[k] * ww[k];
+          const float a = d[k] * wv[s * PV + k];
           dyv[s * PV + k] = d[k];
           xh[s * PV + k] = xhat;
-          wv[s * PV + k] = ww[k];
           s1 += a;
           s2 += a * xhat;
         }
@@ -233,36 +243,29 @@ __global__ void __launch_bounds__(256) ln_bwd_kernel(
         for (int k = 0; k < PV; ++k) {
           const float a = dyv[s * PV + k] * wv[s * PV + k];
           out[k] = (a - s1 - xh[s * PV + k] * s2) * rstd;
-          atomicAdd(&s_gamma[vi * PV + k], dyv[s * PV + k] * xh[s * PV + k]);
-          atomicAdd(&s_beta[vi * PV + k], dyv[s * PV + k]);
+          dg[s * PV + k] += dyv[s * PV + k] * xh[s * PV + k];
+          db[s * PV + k] += dyv[s * PV + k];
         }
         reinterpret_cast<Vec*>(dxr)[vi] = IO::pack(out);
       }
     }
   }
 
-  __syncthreads();
-  float* wg = ws_gamma + (long)blockIdx.x * COLS;
-  float* wb = ws_beta + (long)blockIdx.x * COLS;
-  for (int i = threadIdx.x; i < COLS; i += blockDim.x) {
-    wg[i] = s_gamma[i];
-    wb[i] = s_beta[i];
+  // flush per-wave partials (one slot per wave, no atomics)
+  const long slot = (long)blockIdx.x * 4 + wave;
+  float* wg = ws_gamma + slot * COLS;
+  float* wb = ws_beta + slot * COLS;
+#pragma unroll
+  for (int s = 0; s < RS::kChunks; ++s) {
+    const int vi = s * 64 + lane;
+    if (vi < RS::kVecs) {
+#pragma unroll
+      for (int k = 0; k < PV; ++k) {
+        wg[vi * PV + k] = dg[s * PV + k];
+        wb[vi * PV + k] = db[s * PV + k];
+      }
+    }
   }
-}
-
-template <typename T>
-__global__ void __launch_bounds__(256) ln_bwd_reduce_kernel(
-    const float* __restrict__ ws_gamma, const float* __restrict__ ws_beta,
-    T* __restrict__ dgamma, T* __restrict__ dbeta, int nblocks, int cols) {
-  const int col = blockIdx.x * blockDim.x + threadIdx.x;
-  if (col >= cols) return;
-  float g = 0.f, b = 0.f;
-  for (int i = 0; i < nblocks; ++i) {
-    g += ws_gamma[(long)i * cols + col];
-    b += ws_beta[(long)i * cols + col];
-  }
-  dgamma[col] = (T)g;
-  dbeta[col] = (T)b;
 }
 
 // ---------------------------------------------------------------------------
@@ -302,9 +305,8 @@ static void ln_fwd_dispatch(const void* x, const void* w, const void* b,
 template <typename T>
 static void ln_bwd_dispatch(const void* dy, const void* x, const void* w,
                             const float* mean, const float* rstd, void* dx,
-                            float* ws_gamma, float* ws_beta, void* dgamma,
-                            void* dbeta, long rows, long cols, int nblocks,
-                            hipStream_t s) {
+                            float* ws_gamma, float* ws_beta, long rows,
+                            long cols, int nblocks, hipStream_t s) {
   const dim3 block(256);
 #define LN_BWD_CASE(C)                                                      \
   if (cols == C) {                                                          \
@@ -314,9 +316,6 @@ static void ln_bwd_dispatch(const void* dy, const void* x, const void* w,
   }
   LN_COLS_LIST(LN_BWD_CASE)
 #undef LN_BWD_CASE
-  const dim3 rgrid((cols + 255) / 256);
-  hipLaunchKernelGGL((ln_bwd_reduce_kernel<T>), rgrid, block, 0, s, ws_gamma,
-                     ws_beta, (T*)dgamma, (T*)dbeta, nblocks, (int)cols);
 }
 
 void ln_forward(int dtype, const void* x, const void* w, const void* b,
@@ -339,21 +338,20 @@ void ln_forward(int dtype, const void* x, const void* w, const void* b,
 
 void ln_backward(int dtype, const void* dy, const void* x, const void* w,
                  const float* mean, const float* rstd, void* dx,
-                 float* ws_gamma, float* ws_beta, void* dgamma, void* dbeta,
-                 long rows, long cols, int nblocks, hipStream_t stream) {
+                 float* ws_gamma, float* ws_beta, long rows, long cols,
+                 int nblocks, hipStream_t stream) {
   switch ((Dtype)dtype) {
     case Dtype::F32:
       ln_bwd_dispatch<float>(dy, x, w, mean, rstd, dx, ws_gamma, ws_beta,
-                             dgamma, dbeta, rows, cols, nblocks, stream);
+                             rows, cols, nblocks, stream);
       break;
     case Dtype::BF16:
       ln_bwd_dispatch<__hip_bfloat16>(dy, x, w, mean, rstd, dx, ws_gamma,
-                                      ws_beta, dgamma, dbeta, rows, cols,
-                                      nblocks, stream);
+                                      ws_beta, rows, cols, nblocks, stream);
       break;
     case Dtype::F16:
       ln_bwd_dispatch<__half>(dy, x, w, mean, rstd, dx, ws_gamma, ws_beta,
-                              dgamma, dbeta, rows, cols, nblocks, stream);
+                              rows, cols, nblocks, stream);
       break;
   }
 }

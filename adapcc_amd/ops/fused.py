@@ -59,19 +59,20 @@ class _FusedLayerNormFn(torch.autograd.Function):
         rows = x.numel() // cols
         dyc = dy.contiguous()
         dx = torch.empty_like(x)
-        dgamma = torch.empty_like(weight)
-        dbeta = torch.empty_like(weight)
         nblocks = max(1, min(1024, (rows + 3) // 4))
-        ws = torch.empty(2 * nblocks * cols, dtype=torch.float32,
+        nslots = nblocks * 4  # one partial per wave
+        ws = torch.empty(2 * nslots * cols, dtype=torch.float32,
                          device=x.device)
-        wsg = ws[: nblocks * cols]
-        wsb = ws[nblocks * cols:]
+        wsg = ws[: nslots * cols]
+        wsb = ws[nslots * cols:]
         stream = torch.cuda.current_stream(x.device).cuda_stream
         c.ln_bwd(_DT[x.dtype], dyc.data_ptr(), x.data_ptr(),
                  weight.data_ptr(), mean.data_ptr(), rstd.data_ptr(),
                  dx.data_ptr(), wsg.data_ptr(), wsb.data_ptr(),
-                 dgamma.data_ptr(), dbeta.data_ptr(), rows, cols, nblocks,
-                 stream)
+                 rows, cols, nblocks, stream)
+        # fold the per-wave partials (torch's tree reduction is optimal here)
+        dgamma = wsg.view(nslots, cols).sum(0).to(weight.dtype)
+        dbeta = wsb.view(nslots, cols).sum(0).to(weight.dtype)
         return dx, dgamma, dbeta, None
 
 

@@ -194,12 +194,14 @@ PYBIND11_MODULE(_core, m) {
       .def("ipc_handle",
            [](Engine& e) { return py::bytes(e.ipc_handle()); })
       .def("connect",
-           [](Engine& e, const std::vector<py::bytes>& handles) {
+           [](Engine& e, const std::vector<py::bytes>& handles,
+              const std::vector<int>& peer_devices) {
              std::vector<std::string> hs;
              hs.reserve(handles.size());
              for (const auto& h : handles) hs.push_back(std::string(h));
-             e.connect(hs);
-           })
+             e.connect(hs, peer_devices);
+           },
+           py::arg("handles"), py::arg("peer_devices") = std::vector<int>{})
       .def("set_strategy", &Engine::set_strategy, py::arg("parents"),
            py::arg("chunk_bytes"))
       .def("connect_local", &Engine::connect_local, py::arg("peer_addrs"))

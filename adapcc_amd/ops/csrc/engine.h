@@ -23,7 +23,8 @@ class Engine {
   Engine& operator=(const Engine&) = delete;
 
   std::string ipc_handle() const;
-  void connect(const std::vector<std::string>& handles);
+  void connect(const std::vector<std::string>& handles,
+               const std::vector<int>& peer_devices = {});
   // Test/emulation path: wire peers by raw device address (same process,
   // same device). Exercises the full flag protocol on one GPU.
   void connect_local(const std::vector<uintptr_t>& peer_addrs);

@@ -118,10 +118,28 @@ std::string Engine::ipc_handle() const {
   return std::string(reinterpret_cast<const char*>(&h), sizeof(h));
 }
 
-void Engine::connect(const std::vector<std::string>& handles) {
+void Engine::connect(const std::vector<std::string>& handles,
+                     const std::vector<int>& peer_devices) {
   if ((int)handles.size() != world_)
     throw std::runtime_error("connect: need one handle per rank");
   HIP_CHECK(hipSetDevice(device_));
+  // Belt-and-braces for the multi-GPU case: enable peer access to every
+  // distinct peer device before opening their IPC handles (the lazy flag
+  // on hipIpcOpenMemHandle should do this on demand; being explicit makes
+  // a misconfigured xGMI link fail loudly at setup, caught by self_test).
+  for (int dev : peer_devices) {
+    if (dev < 0 || dev == device_) continue;
+    int can = 0;
+    if (hipDeviceCanAccessPeer(&can, device_, dev) == hipSuccess && can) {
+      hipError_t e = hipDeviceEnablePeerAccess(dev, 0);
+      if (e != hipSuccess && e != hipErrorPeerAccessAlreadyEnabled) {
+        throw std::runtime_error(std::string("enable peer access to dev ") +
+                                 std::to_string(dev) + ": " +
+                                 hipGetErrorString(e));
+      }
+      (void)hipGetLastError();  // clear AlreadyEnabled sticky state
+    }
+  }
   for (int r = 0; r < world_; ++r) {
     if (r == rank_) continue;
     if (handles[r].size() != sizeof(hipIpcMemHandle_t))

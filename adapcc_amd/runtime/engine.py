@@ -100,9 +100,11 @@ class NativeEngine:
         import torch.distributed as dist
 
         handle = self._eng.ipc_handle()
-        gathered: List[Optional[bytes]] = [None] * self.world_size
-        dist.all_gather_object(gathered, handle, group=group)
-        self._eng.connect(list(gathered))
+        gathered: List[Optional[tuple]] = [None] * self.world_size
+        dist.all_gather_object(gathered, (handle, self.device), group=group)
+        handles = [g[0] for g in gathered]
+        devices = [g[1] for g in gathered]
+        self._eng.connect(handles, devices)
         self._connected = True
 
     def set_strategy(self, strategy: Strategy) -> None:

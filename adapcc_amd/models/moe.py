@@ -13,7 +13,6 @@ make the exchange a single equal-split all-to-all in each direction.
 from __future__ import annotations
 
 import math
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -3,8 +3,6 @@ kernels in csrc/lnorm.hip)."""
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.nn as nn
 

@@ -18,7 +18,7 @@ Arc-file format (reference's): one ``src dst capacity`` per line.
 
 from __future__ import annotations
 
-from typing import Dict, List, Sequence, Tuple
+from typing import List, Sequence, Tuple
 
 import numpy as np
 from scipy.optimize import linprog

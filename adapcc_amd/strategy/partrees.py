@@ -23,8 +23,7 @@ node expanded into an intra-node chain of its local GPUs (trees.py:45-65).
 
 from __future__ import annotations
 
-import math
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional
 
 from ..topology.formats import (
     LogicalGraph,

@@ -17,7 +17,6 @@ Implemented with stdlib ElementTree (the reference used tinyxml2 + xmltodict).
 
 from __future__ import annotations
 
-import io
 import os
 import xml.etree.ElementTree as ET
 from dataclasses import dataclass, field

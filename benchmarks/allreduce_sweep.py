@@ -161,7 +161,6 @@ def main():
                 continue
             from adapcc_amd.runtime.engine import NativeEngine
             from adapcc_amd.strategy.partrees import synthesize_stars
-            from adapcc_amd.topology.formats import Strategy
 
             eng = NativeEngine(rank, world, device=device.index,
                                cap_bytes=max(sizes) + (1 << 20))

@@ -19,7 +19,6 @@ import io
 import os
 import sys
 import tempfile
-import time
 
 import torch
 import torch.distributed as dist

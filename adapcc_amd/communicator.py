@@ -177,6 +177,8 @@ class Communicator:
                 self._setup_pg()
         elif transport == "pg":
             self._setup_pg()
+        elif transport == "p2p":
+            self._setup_p2p()
         else:
             raise ValueError(f"unknown ADAPCC_TRANSPORT={transport}")
         if self.args.relay and self.world_size > 1:
@@ -241,6 +243,16 @@ class Communicator:
         self.engine.set_strategy(self.strategy)
         self.engine.self_test()
         self.effective_transport = "native"
+
+    def _setup_p2p(self) -> None:
+        from .runtime.p2p_engine import P2PTreeEngine
+
+        self.engine = P2PTreeEngine(self.rank, self.world_size,
+                                    group=self.group)
+        self.engine.bootstrap(group=self.group)
+        self.engine.set_strategy(self.strategy)
+        self.engine.self_test()
+        self.effective_transport = "p2p"
 
     def _setup_pg(self) -> None:
         from .runtime.fallback import ProcessGroupEngine

@@ -18,6 +18,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from ..ops.fused import FusedLayerNorm
+
 
 class _AllToAll(torch.autograd.Function):
     """Differentiable equal-split all-to-all; the backward pass is the
@@ -135,9 +137,9 @@ class MoETransformerBlock(nn.Module):
 
     def __init__(self, d_model: int = 1024, n_head: int = 8, **moe_kw):
         super().__init__()
-        self.ln1 = nn.LayerNorm(d_model)
+        self.ln1 = FusedLayerNorm(d_model)
         self.attn = nn.MultiheadAttention(d_model, n_head, batch_first=True)
-        self.ln2 = nn.LayerNorm(d_model)
+        self.ln2 = FusedLayerNorm(d_model)
         self.moe = MoEMLP(d_model=d_model, **moe_kw)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

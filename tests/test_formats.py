@@ -101,3 +101,15 @@ def test_profile_roundtrip(tmp_path):
     prof2 = load_profile(p)
     assert prof2.latency[(0, 1)] == pytest.approx(12.5)
     assert prof2.bandwidth[(0, 1)] == pytest.approx(150.0)
+
+
+def test_shipped_strategy_files():
+    import glob
+    import os
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    files = glob.glob(os.path.join(repo, "strategy", "*.xml"))
+    assert files, "shipped strategy examples missing"
+    for f in files:
+        s = load_strategy(f)
+        s.validate(len(s.ranks()))

@@ -361,3 +361,63 @@ def _oversized_tensor_split(rank, world):
 
 def test_oversized_tensor_split():
     assert all(run_mp(_oversized_tensor_split, 2, backend="gloo", timeout=180))
+
+
+def _gpt2_ddp_rehearsal(rank, world):
+    """Closest 1-GPU rehearsal of the driver's multi-GPU bench: GPT-2 tiny,
+    DDP + adapcc hook over the NATIVE engine, bf16-master optimizer, 3
+    steps; replicas must remain byte-consistent."""
+    os.environ["ADAPCC_TIMEOUT_MS"] = "30000"
+    os.environ["ADAPCC_TRANSPORT"] = "native"
+    import torch
+    import torch.distributed as dist
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    torch.cuda.set_device(0)
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.models.gpt2 import GPT2, GPT2Config
+    from adapcc_amd.runtime.hook import AdapccDDPState, adapcc_allreduce_hook
+
+    AdapCC.init(CommArgs(entry_point=-1), 0, rank, world)
+    AdapCC.setup()
+    torch.manual_seed(7)
+    model = GPT2(GPT2Config.tiny()).cuda().to(torch.bfloat16)
+    ddp = DDP(model, bucket_cap_mb=4)
+    state = AdapccDDPState(AdapCC.communicator)
+    ddp.register_comm_hook(state, adapcc_allreduce_hook)
+
+    params = [p for p in ddp.parameters() if p.requires_grad]
+    masters = [p.detach().float().clone() for p in params]
+    for m in masters:
+        m.grad = torch.zeros_like(m)
+    opt = torch.optim.AdamW(masters, lr=1e-3, fused=True)
+
+    torch.manual_seed(100 + rank)
+    x = torch.randint(0, 2048, (2, 128), device="cuda")
+    losses = []
+    for step in range(3):
+        state.on_step(step)
+        for p in params:
+            p.grad = None
+        _, loss = ddp(x, x)
+        loss.backward()
+        torch._foreach_copy_([m.grad for m in masters],
+                             [p.grad for p in params])
+        opt.step()
+        with torch.no_grad():
+            torch._foreach_copy_(params, masters)
+        losses.append(float(loss))
+    torch.cuda.synchronize()
+    assert all(map(lambda v: v == v, losses)), losses  # finite
+
+    flat = torch.cat([p.detach().float().flatten() for p in params]).cpu()
+    g = [torch.zeros_like(flat) for _ in range(world)]
+    dist.all_gather(g, flat)
+    for other in g:
+        assert torch.allclose(other, g[0], atol=1e-5)
+    AdapCC.clear()
+    return True
+
+
+def test_gpt2_ddp_rehearsal():
+    assert all(run_mp(_gpt2_ddp_rehearsal, 2, backend="gloo", timeout=300))

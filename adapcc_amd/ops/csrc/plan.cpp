@@ -283,6 +283,7 @@ PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
                            long total_elems, int esize, long chunk_bytes,
                            uint64_t active_mask) {
   const int world = shape.world;
+  if (root < 0 || root >= world) throw std::runtime_error("reduce: bad root");
   const int T = (int)shape.parents.size();
   auto active = mask_to_active(active_mask, world);
 
@@ -328,6 +329,7 @@ PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
 
 PlanData build_broadcast_plan(int world, int rank, int root, long total_elems,
                               int esize, long chunk_bytes) {
+  if (root < 0 || root >= world) throw std::runtime_error("broadcast: bad root");
   const int T = std::min<long>(std::min(world, 8), kMaxTrees);
   const long per_raw = (total_elems + T - 1) / T;
   const long per = (per_raw + kAlignE - 1) / kAlignE * kAlignE;

@@ -50,7 +50,6 @@ def adapcc_allreduce_hook(
             state.comm.notify_hook_ready(state.step)
         if state.step == 1 and hasattr(bucket, "index"):
             pass  # bucket layout is stable from step 1 on; nothing to record
-    n_active = (len(state.active) if state.active else state.comm.world_size)
     state.comm.all_reduce(tensor, active=state.active, average=True)
     fut: torch.futures.Future = torch.futures.Future()
     fut.set_result(tensor)

@@ -22,6 +22,9 @@ import os
 import sys
 import time
 
+import os as _os
+_os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
 import torch
 import torch.distributed as dist
 from torch.nn.parallel import DistributedDataParallel as DDP

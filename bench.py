@@ -1,10 +1,11 @@
 #!/usr/bin/env python3
 """Flagship benchmark: GPT-2 small DDP training step over the adapcc engine.
 
-Measures whole-job tokens/sec (bf16 autocast, synthetic tokens, random-init
-weights) with gradient buckets allreduced by the adapcc_amd engine via the
-DDP communication hook — the reference's train_ddp.py workload re-targeted
-at GPT-2 small per BASELINE.json.
+Measures whole-job tokens/sec (bf16 compute with fp32-master AdamW by
+default, synthetic tokens, random-init weights) with gradient buckets
+allreduced by the adapcc_amd engine via the DDP communication hook — the
+reference's train_ddp.py workload re-targeted at GPT-2 small per
+BASELINE.json.
 
 Launch (driver contract):
     python bench.py --gpus 1 --steps K --warmup W

@@ -26,6 +26,11 @@ class AdapccDDPState:
     comm: Communicator
     step: int = 0
     bucket_elems: List[int] = field(default_factory=list)
+    # BSP mode (reference commu.py:424-426): an inactive (straggler) rank
+    # keeps its OWN gradients for the step instead of adopting the active
+    # set's average. Default off: the engine already delivers the reduced
+    # result to inactive ranks, which keeps replicas consistent.
+    bsp_mode: bool = False
     _first_bucket_of_step: bool = True
 
     def on_step(self, step: int) -> None:
@@ -50,7 +55,15 @@ def adapcc_allreduce_hook(
             state.comm.notify_hook_ready(state.step)
         if state.step == 1 and hasattr(bucket, "index"):
             pass  # bucket layout is stable from step 1 on; nothing to record
-    state.comm.all_reduce(tensor, active=state.active, average=True)
+    active = state.active
+    inactive_bsp = (state.bsp_mode and active is not None
+                    and state.comm.rank not in active)
+    saved = tensor.clone() if inactive_bsp else None
+    state.comm.all_reduce(tensor, active=active, average=True)
+    if inactive_bsp:
+        # keep the straggler's local gradients (the collective still ran so
+        # peers were not blocked — reference BSP semantics)
+        tensor.copy_(saved)
     fut: torch.futures.Future = torch.futures.Future()
     fut.set_result(tensor)
     return fut

@@ -241,3 +241,46 @@ def _other_primitives(rank, world):
 
 def test_other_primitives():
     assert all(run_mp(_other_primitives, 2))
+
+
+def _bsp_mode(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    import torch.nn as nn
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.runtime.hook import AdapccDDPState, adapcc_allreduce_hook
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+    comm = AdapCC.communicator
+    comm.active_ranks = [0]  # rank 1 is an inactive straggler
+
+    torch.manual_seed(7)
+    model = nn.Linear(8, 4)
+    ddp = DDP(model, bucket_cap_mb=1)
+    state = AdapccDDPState(comm, bsp_mode=True)
+    ddp.register_comm_hook(state, adapcc_allreduce_hook)
+    state.on_step(2)
+
+    torch.manual_seed(50 + rank)
+    x = torch.randn(4, 8)
+    ((ddp(x)) ** 2).mean().backward()
+    got = torch.cat([p.grad.flatten() for p in ddp.parameters()])
+
+    # local-only reference gradients for this rank
+    torch.manual_seed(7)
+    ref = nn.Linear(8, 4)
+    ((ref(x)) ** 2).mean().backward()
+    local = torch.cat([p.grad.flatten() for p in ref.parameters()])
+
+    if rank == 1:  # BSP: straggler keeps its own gradients
+        assert torch.allclose(got, local, atol=1e-6)
+    else:  # active set of one: average == its local grads too
+        assert torch.allclose(got, local, atol=1e-6)
+    AdapCC.clear()
+    return True
+
+
+def test_bsp_mode():
+    assert all(run_mp(_bsp_mode, 2, backend="gloo", timeout=120))

@@ -102,3 +102,44 @@ class AdapCC:
     def _require_init(cls) -> None:
         if cls.communicator is None:
             raise RuntimeError("AdapCC.init() has not been called")
+
+
+def _primitive_demo() -> None:
+    """Golden-output primitive run (reference: adapcc.py:81-117 __main__ and
+    log/primitive): each rank allreduces a ones*(rank+1) 16-float tensor and
+    prints the world-sum tensor.
+
+        python -m torch.distributed.run --nproc-per-node 4 \
+            --master-addr 127.0.0.1 -m adapcc_amd.adapcc
+    """
+    import os
+
+    import torch.distributed as dist
+
+    rank = int(os.environ.get("RANK", 0))
+    local_rank = int(os.environ.get("LOCAL_RANK", 0))
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank % max(1, torch.cuda.device_count()))
+    if world > 1:
+        backend = ("nccl" if use_cuda and torch.cuda.device_count() >= world
+                   else "gloo")
+        dist.init_process_group(backend)
+
+    AdapCC.init(CommArgs(entry_point=-1), local_rank, rank, world)
+    AdapCC.setup(Primitive.ALLREDUCE)
+    device = "cuda" if use_cuda else "cpu"
+    for it in range(2):
+        t = torch.full((16,), float(rank + 1), device=device)
+        AdapCC.allreduce(t)
+        AdapCC.communicator.synchronize()
+        print(f"[Rank {rank}] iter {it}: {t.cpu().tolist()[:4]} ... "
+              f"(expect {sum(range(1, world + 1)) * (1.0)})", flush=True)
+    AdapCC.clear()
+    if world > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    _primitive_demo()

@@ -293,9 +293,9 @@ class Communicator:
         if self.engine is not None:
             self.engine.synchronize()
 
-    # The remaining primitives currently route through the native engine when
-    # it implements them, else torch.distributed (RCCL/gloo). The tree engine
-    # gains native reduce/broadcast/allgather/alltoall incrementally.
+    # Primitive dispatch: the native engine implements all six primitives;
+    # transports without a native method (pg fallback, p2p tier off the
+    # allreduce path) route through torch.distributed.
 
     def reduce(self, tensor: torch.Tensor, root: int = 0,
                active: Optional[Sequence[int]] = None) -> torch.Tensor:

@@ -251,7 +251,8 @@ __global__ void __launch_bounds__(256) ln_bwd_kernel(
     }
   }
 
-  // flush per-wave partials (one slot per wave, no atomics)
+  // flush per-wave partials (one slot per wave, no atomics); each lane's
+  // PV floats are contiguous -> vectorized float4 stores
   const long slot = (long)blockIdx.x * 4 + wave;
   float* wg = ws_gamma + slot * COLS;
   float* wb = ws_beta + slot * COLS;
@@ -260,9 +261,13 @@ __global__ void __launch_bounds__(256) ln_bwd_kernel(
     const int vi = s * 64 + lane;
     if (vi < RS::kVecs) {
 #pragma unroll
-      for (int k = 0; k < PV; ++k) {
-        wg[vi * PV + k] = dg[s * PV + k];
-        wb[vi * PV + k] = db[s * PV + k];
+      for (int q = 0; q < PV / 4; ++q) {
+        reinterpret_cast<float4*>(wg + vi * PV)[q] =
+            make_float4(dg[s * PV + 4 * q], dg[s * PV + 4 * q + 1],
+                        dg[s * PV + 4 * q + 2], dg[s * PV + 4 * q + 3]);
+        reinterpret_cast<float4*>(wb + vi * PV)[q] =
+            make_float4(db[s * PV + 4 * q], db[s * PV + 4 * q + 1],
+                        db[s * PV + 4 * q + 2], db[s * PV + 4 * q + 3]);
       }
     }
   }

@@ -358,6 +358,11 @@ class Communicator:
             except Exception:
                 pass
             self.engine = None
+            # all ranks drop their hipIpc imports/exports before anyone
+            # rebuilds (reconstruct_topology) — dmabuf refcounts make the
+            # teardown order safe, the barrier makes it deterministic
+            if dist.is_initialized() and self.world_size > 1:
+                dist.barrier(group=self.group)
         self._setup_done = False
         if not keep_coordinator:
             if self.controller is not None:

@@ -65,20 +65,97 @@ class P2PTreeEngine:
     ) -> torch.Tensor:
         if self.world_size == 1:
             return tensor
-        if self.parents is None:
-            raise RuntimeError("no strategy set")
-        if not tensor.is_contiguous():
-            raise ValueError("p2p engine requires a contiguous tensor")
+        self._check(tensor)
         act = sorted(active) if active else list(range(self.world_size))
         plan = self.core.compute_plan(
             self.parents, self.rank, tensor.numel(), tensor.element_size(),
             self.chunk_bytes, act)
-        self._execute(plan, tensor, act, average)
+        self._execute(plan, tensor, tensor, act, average)
         return tensor
 
-    def _execute(self, plan: dict, tensor: torch.Tensor, act: List[int],
+    def reduce(self, tensor: torch.Tensor, root: int = 0,
+               active: Optional[Sequence[int]] = None,
+               average: bool = False) -> torch.Tensor:
+        if self.world_size == 1:
+            return tensor
+        self._check(tensor)
+        act = sorted(active) if active else list(range(self.world_size))
+        plan = self.core.compute_primitive_plan(
+            "reduce", self.world_size, self.rank, tensor.numel(),
+            tensor.element_size(), self.chunk_bytes, root=root,
+            parents=self.parents, active=act)
+        self._execute(plan, tensor, tensor, act, average)
+        return tensor
+
+    def broadcast(self, tensor: torch.Tensor, root: int = 0) -> torch.Tensor:
+        if self.world_size == 1:
+            return tensor
+        self._check(tensor)
+        plan = self.core.compute_primitive_plan(
+            "broadcast", self.world_size, self.rank, tensor.numel(),
+            tensor.element_size(), self.chunk_bytes, root=root)
+        self._execute(plan, tensor, tensor, list(range(self.world_size)),
+                      False)
+        return tensor
+
+    def all_gather(self, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        self._check(tensor)
+        if out.numel() != tensor.numel() * self.world_size:
+            raise ValueError("all_gather: out must be world_size * in")
+        if self.world_size == 1:
+            out.view(-1).copy_(tensor.view(-1))
+            return out
+        plan = self.core.compute_primitive_plan(
+            "allgather", self.world_size, self.rank, tensor.numel(),
+            tensor.element_size(), self.chunk_bytes)
+        self._execute(plan, tensor, out, list(range(self.world_size)), False)
+        return out
+
+    def reduce_scatter(self, out: torch.Tensor, tensor: torch.Tensor,
+                       active: Optional[Sequence[int]] = None,
+                       average: bool = False) -> torch.Tensor:
+        self._check(tensor)
+        if tensor.numel() != out.numel() * self.world_size:
+            raise ValueError("reduce_scatter: in must be world_size * out")
+        if self.world_size == 1:
+            out.view(-1).copy_(tensor.view(-1))
+            return out
+        act = sorted(active) if active else list(range(self.world_size))
+        plan = self.core.compute_primitive_plan(
+            "reducescatter", self.world_size, self.rank, out.numel(),
+            tensor.element_size(), self.chunk_bytes, active=act)
+        self._execute(plan, tensor, out, act, average)
+        return out
+
+    def all_to_all(self, out: torch.Tensor, tensor: torch.Tensor) -> torch.Tensor:
+        self._check(tensor)
+        if out.numel() != tensor.numel():
+            raise ValueError("all_to_all: out must match in size")
+        if out.data_ptr() == tensor.data_ptr():
+            raise ValueError("all_to_all: out must not alias the input")
+        if tensor.numel() % self.world_size:
+            raise ValueError("all_to_all: size must divide world_size")
+        if self.world_size == 1:
+            out.view(-1).copy_(tensor.view(-1))
+            return out
+        plan = self.core.compute_primitive_plan(
+            "alltoall", self.world_size, self.rank,
+            tensor.numel() // self.world_size, tensor.element_size(),
+            self.chunk_bytes)
+        self._execute(plan, tensor, out, list(range(self.world_size)), False)
+        return out
+
+    def _check(self, tensor: torch.Tensor) -> None:
+        if self.parents is None:
+            raise RuntimeError("no strategy set")
+        if not tensor.is_contiguous():
+            raise ValueError("p2p engine requires a contiguous tensor")
+
+    def _execute(self, plan: dict, in_tensor: torch.Tensor,
+                 out_tensor: torch.Tensor, act: List[int],
                  average: bool) -> None:
-        flat = tensor.view(-1)
+        flat = in_tensor.view(-1)
+        out_flat = out_tensor.view(-1)
         # staging buffers per (tree, chunk): received partials / results
         acc: Dict[Tuple[int, int], torch.Tensor] = {}
         result: Dict[Tuple[int, int], torch.Tensor] = {}
@@ -104,11 +181,13 @@ class P2PTreeEngine:
         # reduce receives (the device engine's copyin-before-reduce order)
         if me_active:
             for u in plan["copy"]:
-                if u["notify_to"] and u["flag_space"] == 0:
-                    t, c = u["tree"], u["chunk"]
-                    off, n = u["offset"], u["count"]
-                    for dst in u["notify_to"]:
-                        push(dst, t, c, flat[off:off + n], phase=0)
+                if not u["notify_to"]:
+                    continue
+                t, c = u["tree"], u["chunk"]
+                off, n = u["offset"], u["count"]
+                phase = 0 if u["flag_space"] == 0 else 2
+                for dst in u["notify_to"]:
+                    push(dst, t, c, flat[off:off + n], phase=phase)
 
         # 2) reduce phase in global (chunk, tree) order
         for u in plan["reduce"]:
@@ -130,13 +209,13 @@ class P2PTreeEngine:
                 for k in u["publish_to"]:
                     push(k, t, c, out, phase=2)
 
-        # 3) broadcast phase: receive results, write out, forward
+        # 3) broadcast/receive phase: pull published chunks into OUT
         for u in plan["bcast"]:
             t, c = u["tree"], u["chunk"]
             soff, doff, n = u["src_offset"], u["dst_offset"], u["count"]
             if u["parent"] < 0:
                 buf = acc.get((t, c))
-                if buf is None:  # self-staged (e.g. broadcast root)
+                if buf is None:  # self-staged source (allgather/alltoall/bcast)
                     buf = flat[soff:soff + n]
             else:
                 buf = pull(u["parent"], t, c, n, phase=2)
@@ -145,9 +224,9 @@ class P2PTreeEngine:
                 for k in u["publish_to"]:
                     push(k, t, c, buf, phase=2)
             if scale != 1.0:
-                flat[doff:doff + n] = buf * scale
+                out_flat[doff:doff + n] = buf * scale
             else:
-                flat[doff:doff + n] = buf
+                out_flat[doff:doff + n] = buf
 
         for w in pending:
             w.wait()

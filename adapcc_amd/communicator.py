@@ -51,13 +51,17 @@ class CommArgs:
     chunk_bytes: int = 2 * 1024 * 1024
     relay: bool = False       # straggler-adaptive active sets via coordinator
     coordinator_port: int = 50051
+    # size-adaptive transport: tensors below this many bytes take the RCCL
+    # collective (latency-optimal); larger ones take the tree engine
+    # (bandwidth-optimal star forest). 0 disables the hybrid.
+    small_threshold: int = 64 * 1024
 
     @classmethod
     def from_namespace(cls, ns) -> "CommArgs":
         kw = {}
         for f in ("port", "entry_point", "strategy_file", "logical_graph",
                   "parallel_degree", "profile_freq", "policy", "chunk_bytes",
-                  "relay", "coordinator_port"):
+                  "relay", "coordinator_port", "small_threshold"):
             if hasattr(ns, f) and getattr(ns, f) is not None:
                 kw[f] = getattr(ns, f)
         return cls(**kw)
@@ -279,9 +283,30 @@ class Communicator:
         from .utils.metrics import GLOBAL as metrics
 
         metrics.inc("allreduce_calls")
-        metrics.inc("allreduce_bytes", tensor.numel() * tensor.element_size())
+        nbytes = tensor.numel() * tensor.element_size()
+        metrics.inc("allreduce_bytes", nbytes)
         if active is not None:
             metrics.inc("allreduce_relay_calls")
+        # size-adaptive transport: latency-bound small tensors go straight
+        # to the RCCL/gloo collective; the tree engine owns the
+        # bandwidth-bound regime
+        if (
+            self.args.small_threshold
+            and nbytes < self.args.small_threshold
+            and self.effective_transport in ("native", "p2p")
+            and self.world_size > 1
+            and dist.is_initialized()
+        ):
+            metrics.inc("allreduce_small_pg_calls")
+            n = self.world_size
+            if active is not None and len(active) > 0 and                     len(set(active)) < self.world_size:
+                n = len(set(active))
+                if self.rank not in set(active):
+                    tensor.zero_()
+            dist.all_reduce(tensor, group=self.group)
+            if average:
+                tensor.div_(n)
+            return tensor
         return self.engine.all_reduce(tensor, active=active, average=average)
 
     def stats(self) -> dict:

@@ -284,3 +284,30 @@ def _bsp_mode(rank, world):
 
 def test_bsp_mode():
     assert all(run_mp(_bsp_mode, 2, backend="gloo", timeout=120))
+
+
+def _hybrid_threshold(rank, world):
+    """Size-adaptive transport: small tensors route to the pg collective,
+    large ones to the engine; results identical either way."""
+    os.environ["ADAPCC_TRANSPORT"] = "p2p"  # engine path observable on CPU
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.utils.metrics import GLOBAL as metrics
+
+    AdapCC.init(CommArgs(entry_point=-1, small_threshold=1024), rank, rank,
+                world)
+    AdapCC.setup()
+    small = torch.full((16,), float(rank + 1))       # < 1 KB -> pg
+    big = torch.full((4096,), float(rank + 1))       # > 1 KB -> engine
+    AdapCC.allreduce(small)
+    AdapCC.allreduce(big)
+    expect = float(sum(range(1, world + 1)))
+    assert torch.allclose(small, torch.full_like(small, expect))
+    assert torch.allclose(big, torch.full_like(big, expect))
+    snap = metrics.snapshot()
+    assert snap.get("allreduce_small_pg_calls", 0) == 1, snap
+    AdapCC.clear()
+    return True
+
+
+def test_hybrid_small_threshold():
+    assert all(run_mp(_hybrid_threshold, 2, backend="gloo", timeout=120))

@@ -229,10 +229,17 @@ class Communicator:
 
     def notify_hook_ready(self, step: int) -> None:
         """First DDP bucket of a step: negotiate the straggler-adaptive
-        active set (reference: commu.py:387-394 + rpc_server hook_fetch)."""
+        active set (reference: commu.py:387-394 + rpc_server hook_fetch).
+        Negotiation latency is tracked per call (reference measured
+        ~0.8-1.9 ms per step, proto/latency_0.0.txt)."""
         if self.hooker is None:
             return
+        from .utils.metrics import GLOBAL as metrics
+
+        metrics.timer_start("relay_negotiation")
         active = self.hooker.send_ready_request(step)
+        metrics.timer_stop("relay_negotiation")
+        metrics.inc("relay_negotiations")
         self.active_ranks = (
             None if len(active) >= self.world_size else active
         )

@@ -112,6 +112,13 @@ def main():
         print(f"wait spread ms: min {1000 * min(waits):.2f} "
               f"max {1000 * max(waits):.2f} "
               f"mean {1000 * sum(waits) / len(waits):.2f} -> {args.out}")
+        if args.relay:
+            stats = AdapCC.communicator.stats()
+            n = stats.get("relay_negotiations", 0)
+            if n:
+                ms = 1000 * stats.get("relay_negotiation_s", 0) / n
+                print(f"relay negotiation latency: {ms:.2f} ms/step over "
+                      f"{int(n)} steps")
 
     AdapCC.clear()
     dist.destroy_process_group()

@@ -321,3 +321,62 @@ def test_alltoall(world):
         expect = np.concatenate([user[s][r * L:(r + 1) * L]
                                  for s in range(world)])
         np.testing.assert_allclose(sim.out[r], expect, rtol=1e-6, atol=1e-6)
+
+
+# ---------------------------------------------------------------------------
+# additional edge coverage
+# ---------------------------------------------------------------------------
+
+
+def test_allreduce_exact_capacity_chunks():
+    # slice == exactly kMaxChunkSlots chunks boundary: chunk auto-growth
+    world = 2
+    total = 2 * 64 * 600  # forces > 512 chunks at chunk_bytes=256 -> growth
+    sim_allreduce(stars(world), total=total, chunk_bytes=256)
+
+
+def test_allgather_multi_chunk():
+    world = 4
+    L = 3000
+    plans = prim_plans("allgather", world, L, chunk_bytes=1024)
+    user = rand_inputs(world, L)
+    sim = Sim(world, plans, user, world * L)
+    sim.run()
+    expect = np.concatenate(user)
+    for r in range(world):
+        np.testing.assert_allclose(sim.out[r], expect, rtol=1e-6, atol=1e-6)
+
+
+def test_alltoall_multi_chunk_slots():
+    world = 4
+    L = 2000
+    plans = prim_plans("alltoall", world, L, chunk_bytes=512)
+    user = rand_inputs(world, world * L)
+    sim = Sim(world, plans, user, world * L)
+    sim.run()
+    for r in range(world):
+        expect = np.concatenate([user[s][r * L:(r + 1) * L]
+                                 for s in range(world)])
+        np.testing.assert_allclose(sim.out[r], expect, rtol=1e-6, atol=1e-6)
+
+
+def test_reducescatter_relay_subset():
+    world = 4
+    L = 256
+    active = [0, 3]
+    plans = prim_plans("reducescatter", world, L, active=active)
+    user = rand_inputs(world, world * L)
+    sim = Sim(world, plans, user, L)
+    sim.run()
+    total = np.sum([user[r] for r in active], axis=0)
+    for r in range(world):
+        np.testing.assert_allclose(sim.out[r], total[r * L:(r + 1) * L],
+                                   rtol=1e-5, atol=1e-5)
+
+
+def test_single_element_tensor():
+    sim_allreduce(stars(4), total=1)
+
+
+def test_sixteen_ranks():
+    sim_allreduce(stars(16), total=2048)

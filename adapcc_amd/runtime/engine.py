@@ -139,6 +139,9 @@ class NativeEngine:
         active_list = list(active) if active is not None else []
         stream = torch.cuda.current_stream(tensor.device).cuda_stream
 
+        from ..utils.metrics import GLOBAL as metrics
+
+        metrics.inc("native_allreduce_calls")
         max_elems = self.cap_bytes // esize
         numel = tensor.numel()
         if numel <= max_elems:

@@ -247,12 +247,34 @@ class Communicator:
     def _setup_native(self) -> None:
         from .runtime.engine import NativeEngine
 
-        torch.cuda.set_device(self.local_rank)
-        self.engine = NativeEngine(self.rank, self.world_size,
-                                   device=self.local_rank)
-        self.engine.bootstrap(group=self.group)
-        self.engine.set_strategy(self.strategy)
-        self.engine.self_test()
+        # Phase 1: local construction. Failures here (e.g. not enough
+        # devices) must be agreed on COLLECTIVELY before any engine
+        # collective runs, or ranks desynchronize into a cross-transport
+        # deadlock (one rank in the native bootstrap, another fallen back).
+        ndev = max(1, torch.cuda.device_count())
+        dev = self.local_rank % ndev
+        err: Optional[Exception] = None
+        engine = None
+        try:
+            torch.cuda.set_device(dev)
+            engine = NativeEngine(self.rank, self.world_size, device=dev)
+        except Exception as e:  # noqa: BLE001 - agreed on below
+            err = e
+        if self.world_size > 1 and dist.is_initialized():
+            oks: List[Optional[bool]] = [None] * self.world_size
+            dist.all_gather_object(oks, err is None, group=self.group)
+            if not all(oks):
+                del engine
+                raise RuntimeError(
+                    f"native engine unavailable on some rank "
+                    f"(local error: {err})")
+        elif err is not None:
+            raise err
+        # Phase 2: collective bootstrap + strategy + canary.
+        engine.bootstrap(group=self.group)
+        engine.set_strategy(self.strategy)
+        engine.self_test()
+        self.engine = engine
         self.effective_transport = "native"
 
     def _setup_p2p(self) -> None:

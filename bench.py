@@ -66,7 +66,8 @@ def main() -> None:
     world = int(os.environ.get("WORLD_SIZE", 1))
 
     use_cuda = torch.cuda.is_available() and args.device != "cpu"
-    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    dev_idx = local_rank % max(1, torch.cuda.device_count() or 1)
+    device = torch.device(f"cuda:{dev_idx}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
 

@@ -102,7 +102,8 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     use_cuda = torch.cuda.is_available()
-    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+    dev_idx = local_rank % max(1, torch.cuda.device_count() or 1)
+    device = torch.device(f"cuda:{dev_idx}" if use_cuda else "cpu")
     if use_cuda:
         torch.cuda.set_device(device)
     if world > 1:

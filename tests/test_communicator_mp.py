@@ -311,3 +311,60 @@ def _hybrid_threshold(rank, world):
 
 def test_hybrid_small_threshold():
     assert all(run_mp(_hybrid_threshold, 2, backend="gloo", timeout=120))
+
+
+def _bf16_compress_hook(rank, world):
+    os.environ["ADAPCC_TRANSPORT"] = "pg"
+    import torch.nn as nn
+    from torch.nn.parallel import DistributedDataParallel as DDP
+
+    from adapcc_amd import AdapCC, CommArgs
+    from adapcc_amd.runtime.hook import (AdapccDDPState,
+                                         adapcc_bf16_compress_hook)
+
+    AdapCC.init(CommArgs(entry_point=-1), rank, rank, world)
+    AdapCC.setup()
+    torch.manual_seed(7)
+    model = nn.Linear(64, 32)
+    ddp = DDP(model, bucket_cap_mb=1)
+    state = AdapccDDPState(AdapCC.communicator)
+    ddp.register_comm_hook(state, adapcc_bf16_compress_hook)
+    state.on_step(0)
+    torch.manual_seed(100 + rank)
+    x = torch.randn(8, 64)
+    (ddp(x) ** 2).mean().backward()
+    g1 = torch.cat([p.grad.flatten() for p in ddp.parameters()])
+
+    # fp32 reference
+    torch.manual_seed(7)
+    m2 = nn.Linear(64, 32)
+    ddp2 = DDP(m2, bucket_cap_mb=1)
+    (ddp2(x) ** 2).mean().backward()
+    g2 = torch.cat([p.grad.flatten() for p in ddp2.parameters()])
+    # bf16 wire precision: ~3 decimal digits
+    assert torch.allclose(g1, g2, rtol=2e-2, atol=2e-2), (g1 - g2).abs().max()
+    AdapCC.clear()
+    return True
+
+
+def test_bf16_compress_hook():
+    assert all(run_mp(_bf16_compress_hook, 2, backend="gloo", timeout=120))
+
+
+def _gns_from_ranks(rank, world):
+    import torch.nn as nn
+
+    from adapcc_amd.utils.gns import GNS
+
+    torch.manual_seed(7)
+    model = nn.Linear(16, 4)
+    torch.manual_seed(100 + rank)
+    x = torch.randn(8, 16)
+    (model(x) ** 2).mean().backward()
+    gns = GNS().compute_gns_from_ranks(model, world, per_rank_batch=8)
+    assert gns == gns  # finite, not NaN
+    return True
+
+
+def test_gns_from_ranks():
+    assert all(run_mp(_gns_from_ranks, 2, backend="gloo", timeout=120))

@@ -105,3 +105,22 @@ def test_template_scripts_2proc_cpu(script, extra):
                          env=env, cwd=REPO)
     assert res.returncode == 0, res.stdout[-2000:] + res.stderr[-2000:]
     assert "loss" in res.stdout
+
+
+def test_fused_ln_autocast_fallback_cpu():
+    """Under autocast (mixed weight/input dtypes) FusedLayerNorm must take
+    the stock fallback and match nn.LayerNorm exactly."""
+    from adapcc_amd.ops.fused import FusedLayerNorm
+
+    torch.manual_seed(0)
+    fused = FusedLayerNorm(64)
+    ref = torch.nn.LayerNorm(64)
+    with torch.no_grad():
+        ref.weight.copy_(fused.weight)
+        ref.bias.copy_(fused.bias)
+    x = torch.randn(4, 64)
+    with torch.autocast("cpu", dtype=torch.bfloat16):
+        y1 = fused(x)
+        y2 = ref(x)
+    assert y1.dtype == y2.dtype
+    assert torch.allclose(y1.float(), y2.float())

@@ -44,6 +44,11 @@ class AdapccDDPState:
     def active(self) -> Optional[List[int]]:
         return self.comm.active_ranks
 
+    def bucket_info(self) -> List[int]:
+        """Element counts of the stable DDP buckets, recorded at step 1
+        (reference: log/model_bucket_info.txt + commu.py:409-418)."""
+        return list(self.bucket_elems)
+
 
 def adapcc_allreduce_hook(
     state: AdapccDDPState, bucket
@@ -53,8 +58,10 @@ def adapcc_allreduce_hook(
         state._first_bucket_of_step = False
         if hasattr(state.comm, "notify_hook_ready"):
             state.comm.notify_hook_ready(state.step)
-        if state.step == 1 and hasattr(bucket, "index"):
-            pass  # bucket layout is stable from step 1 on; nothing to record
+    if state.step == 1:
+        # bucket layout is stable from DDP's rebuild at iteration 1 on;
+        # record it (reference log/model_bucket_info.txt)
+        state.bucket_elems.append(tensor.numel())
     active = state.active
     inactive_bsp = (state.bsp_mode and active is not None
                     and state.comm.rank not in active)

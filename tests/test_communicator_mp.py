@@ -79,6 +79,10 @@ def _ddp_hook_training(rank, world):
         loss.backward()
         opt.step()
         losses.append(float(loss))
+    # bucket layout recorded at step 1 (model_bucket_info parity)
+    assert state.bucket_info(), "bucket info not recorded"
+    assert sum(state.bucket_info()) == sum(q.numel()
+                                           for q in model.parameters())
     # replicas must stay in sync: compare a parameter hash across ranks
     p = torch.cat([q.flatten() for q in model.parameters()])
     import torch.distributed as dist

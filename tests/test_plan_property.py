@@ -26,7 +26,7 @@ def random_forest(rng: random.Random, world: int, ntrees: int):
     return parents
 
 
-@pytest.mark.parametrize("seed", range(12))
+@pytest.mark.parametrize("seed", range(24))
 def test_random_forests_allreduce(seed):
     rng = random.Random(seed)
     world = rng.choice([2, 3, 4, 5, 8, 11, 16])
@@ -51,7 +51,7 @@ def test_random_forests_allreduce(seed):
         np.testing.assert_allclose(sim.out[r], expect, rtol=1e-5, atol=1e-5)
 
 
-@pytest.mark.parametrize("seed", range(6))
+@pytest.mark.parametrize("seed", range(10))
 def test_random_deep_chains(seed):
     """Degenerate deep trees (max forwarding depth) with random relays."""
     rng = random.Random(1000 + seed)

@@ -105,15 +105,18 @@ class Engine {
   void* region_ = nullptr;
   size_t region_bytes_ = 0;
   size_t inbox_off_ = 0;
+  size_t slot_bytes_ = 0;   // size of one call slot (pipelined mode: 2 slots)
+  int n_slots_ = 1;         // ADAPCC_PIPELINE=1 -> 2 (cross-call overlap)
   void* peer_base_[kMaxRanks];
 
-  DevTables tabs_{};
+  DevTables tabs_[2] = {};
 
   unsigned long long* counters_ = nullptr;
   uint64_t* h_err_ = nullptr;
 
   hipStream_t s_red_{}, s_bcast_{}, s_err_{};
-  hipEvent_t ev_in_{}, ev_sync0_{}, ev_red_{}, ev_barrier_{};
+  hipEvent_t ev_in_{}, ev_sync0_{}, ev_red_{}, ev_bc_{},
+      ev_barrier_[2] = {};
 
   int num_trees_ = 0;
   long chunk_bytes_ = 4 * 1024 * 1024;

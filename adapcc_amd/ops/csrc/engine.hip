@@ -70,20 +70,29 @@ Engine::Engine(int rank, int world, int device, size_t cap_bytes,
       timeout_ms_(timeout_ms) {
   if (world > kMaxRanks) throw std::runtime_error("world > kMaxRanks");
   HIP_CHECK(hipSetDevice(device_));
-  // region: [send][acc][result][inbox]
+  if (const char* s = getenv("ADAPCC_PIPELINE")) n_slots_ = atoi(s) ? 2 : 1;
+  // region: n_slots x ([send][acc][result][inbox]) — slot-alternated calls
+  // overlap (call k's broadcast with call k+1's reduce) when pipelined
   inbox_off_ = 3 * cap_bytes_;
-  region_bytes_ = inbox_off_ + align_up(sizeof(FlagInbox));
+  slot_bytes_ = inbox_off_ + align_up(sizeof(FlagInbox));
+  region_bytes_ = (size_t)n_slots_ * slot_bytes_;
   HIP_CHECK(hipMalloc(&region_, region_bytes_));
   HIP_CHECK(hipMemset(region_, 0, region_bytes_));
-  HIP_CHECK(hipMalloc(&counters_, 3 * kMaxUnits * sizeof(unsigned long long)));
-  HIP_CHECK(hipMemset(counters_, 0, 3 * kMaxUnits * sizeof(unsigned long long)));
+  HIP_CHECK(hipMalloc(&counters_,
+                      (size_t)n_slots_ * 3 * kMaxUnits *
+                          sizeof(unsigned long long)));
+  HIP_CHECK(hipMemset(counters_, 0,
+                      (size_t)n_slots_ * 3 * kMaxUnits *
+                          sizeof(unsigned long long)));
   HIP_CHECK(hipStreamCreateWithFlags(&s_red_, hipStreamNonBlocking));
   HIP_CHECK(hipStreamCreateWithFlags(&s_bcast_, hipStreamNonBlocking));
   HIP_CHECK(hipStreamCreateWithFlags(&s_err_, hipStreamNonBlocking));
   HIP_CHECK(hipEventCreateWithFlags(&ev_in_, hipEventDisableTiming));
   HIP_CHECK(hipEventCreateWithFlags(&ev_sync0_, hipEventDisableTiming));
   HIP_CHECK(hipEventCreateWithFlags(&ev_red_, hipEventDisableTiming));
-  HIP_CHECK(hipEventCreateWithFlags(&ev_barrier_, hipEventDisableTiming));
+  HIP_CHECK(hipEventCreateWithFlags(&ev_bc_, hipEventDisableTiming));
+  HIP_CHECK(hipEventCreateWithFlags(&ev_barrier_[0], hipEventDisableTiming));
+  HIP_CHECK(hipEventCreateWithFlags(&ev_barrier_[1], hipEventDisableTiming));
   HIP_CHECK(hipHostMalloc(&h_err_, 2 * sizeof(uint64_t)));
   for (int r = 0; r < kMaxRanks; ++r) peer_base_[r] = nullptr;
   peer_base_[rank_] = region_;
@@ -109,7 +118,9 @@ Engine::~Engine() {
   (void)hipEventDestroy(ev_in_);
   (void)hipEventDestroy(ev_sync0_);
   (void)hipEventDestroy(ev_red_);
-  (void)hipEventDestroy(ev_barrier_);
+  (void)hipEventDestroy(ev_bc_);
+  (void)hipEventDestroy(ev_barrier_[0]);
+  (void)hipEventDestroy(ev_barrier_[1]);
 }
 
 std::string Engine::ipc_handle() const {
@@ -167,14 +178,16 @@ void Engine::connect_local(const std::vector<uintptr_t>& peer_addrs) {
 }
 
 void Engine::build_tables() {
-  for (int r = 0; r < world_; ++r) {
-    char* base = static_cast<char*>(peer_base_[r]);
-    tabs_.send[r] = base;
-    tabs_.acc[r] = base + cap_bytes_;
-    tabs_.result[r] = base + 2 * cap_bytes_;
-    tabs_.inbox[r] = reinterpret_cast<FlagInbox*>(base + inbox_off_);
+  for (int slot = 0; slot < n_slots_; ++slot) {
+    for (int r = 0; r < world_; ++r) {
+      char* base = static_cast<char*>(peer_base_[r]) + slot * slot_bytes_;
+      tabs_[slot].send[r] = base;
+      tabs_[slot].acc[r] = base + cap_bytes_;
+      tabs_[slot].result[r] = base + 2 * cap_bytes_;
+      tabs_[slot].inbox[r] = reinterpret_cast<FlagInbox*>(base + inbox_off_);
+    }
+    tabs_[slot].counters = counters_;
   }
-  tabs_.counters = counters_;
 }
 
 void Engine::set_strategy(const std::vector<std::vector<int>>& parents,
@@ -278,37 +291,49 @@ void Engine::enqueue(const Plan& plan, const void* in, void* out,
                      CallArgs& args, void* caller_stream) {
   hipStream_t caller = reinterpret_cast<hipStream_t>(caller_stream);
   HIP_CHECK(hipSetDevice(device_));
-  // serialize after previous call + after caller-produced data
-  if (args.seq > 1) HIP_CHECK(hipStreamWaitEvent(s_red_, ev_barrier_, 0));
+  // Slot-alternated calls: call k reuses slot k % n_slots, so it only
+  // serializes behind call k - n_slots (whose end barrier proved every
+  // peer finished with that slot). With n_slots == 2, call k+1's
+  // copy/reduce overlaps call k's broadcast+barrier on the other stream.
+  const int slot = (int)(args.seq % (uint64_t)n_slots_);
+  const DevTables& tabs = tabs_[slot];
+  unsigned long long* ctr = counters_ + (size_t)slot * 3 * kMaxUnits;
+  if (args.seq > (uint64_t)n_slots_)
+    HIP_CHECK(hipStreamWaitEvent(s_red_, ev_barrier_[slot], 0));
   HIP_CHECK(hipEventRecord(ev_in_, caller));
   HIP_CHECK(hipStreamWaitEvent(s_red_, ev_in_, 0));
 
   const size_t cu64 = sizeof(unsigned long long);
   if (!plan.cunits.empty())
-    HIP_CHECK(hipMemsetAsync(counters_, 0, plan.cunits.size() * cu64, s_red_));
+    HIP_CHECK(hipMemsetAsync(ctr, 0, plan.cunits.size() * cu64, s_red_));
   if (!plan.runits.empty())
-    HIP_CHECK(hipMemsetAsync(counters_ + kMaxUnits, 0,
+    HIP_CHECK(hipMemsetAsync(ctr + kMaxUnits, 0,
                              plan.runits.size() * cu64, s_red_));
   if (!plan.bunits.empty())
-    HIP_CHECK(hipMemsetAsync(counters_ + 2 * kMaxUnits, 0,
+    HIP_CHECK(hipMemsetAsync(ctr + 2 * kMaxUnits, 0,
                              plan.bunits.size() * cu64, s_red_));
   HIP_CHECK(hipEventRecord(ev_sync0_, s_red_));
   HIP_CHECK(hipStreamWaitEvent(s_bcast_, ev_sync0_, 0));
 
+  DevTables launch_tabs = tabs;
+  launch_tabs.counters = ctr;
   launch_collective(args.dtype, in, out, plan.d_c, (int)plan.cunits.size(),
                     plan.d_r, (int)plan.runits.size(), plan.d_b,
-                    (int)plan.bunits.size(), tabs_, args, rank_,
-                    counters_ + kMaxUnits, counters_ + 2 * kMaxUnits,
+                    (int)plan.bunits.size(), launch_tabs, args, rank_,
+                    ctr + kMaxUnits, ctr + 2 * kMaxUnits,
                     wgs_per_group_, n_groups_, s_red_, s_bcast_);
   HIP_CHECK(hipGetLastError());
 
+  // caller sees "result ready" (bcast wrote the user tensor); the end
+  // barrier only gates the engine's own next same-slot call on s_red
+  HIP_CHECK(hipEventRecord(ev_bc_, s_bcast_));
   HIP_CHECK(hipEventRecord(ev_red_, s_red_));
   HIP_CHECK(hipStreamWaitEvent(s_bcast_, ev_red_, 0));
-  launch_barrier(tabs_, args, rank_, world_, plan.d_ranks, plan.nranks,
+  launch_barrier(launch_tabs, args, rank_, world_, plan.d_ranks, plan.nranks,
                  s_bcast_);
   HIP_CHECK(hipGetLastError());
-  HIP_CHECK(hipEventRecord(ev_barrier_, s_bcast_));
-  HIP_CHECK(hipStreamWaitEvent(caller, ev_barrier_, 0));
+  HIP_CHECK(hipEventRecord(ev_barrier_[slot], s_bcast_));
+  HIP_CHECK(hipStreamWaitEvent(caller, ev_bc_, 0));
 }
 
 CallArgs Engine::make_args(Dtype dt, RedOp op, float scale, long elems) {
@@ -425,14 +450,22 @@ void Engine::reduce_scatter(const void* in, void* out, long out_elems,
 std::string Engine::dump_inbox() {
   std::string out(sizeof(FlagInbox), '\0');
   HIP_CHECK(hipDeviceSynchronize());
-  HIP_CHECK(hipMemcpy(out.data(), tabs_.inbox[rank_], sizeof(FlagInbox),
+  HIP_CHECK(hipMemcpy(out.data(), tabs_[0].inbox[rank_], sizeof(FlagInbox),
                       hipMemcpyDeviceToHost));
   return out;
 }
 
 std::pair<uint64_t, uint64_t> Engine::query_error() {
   if (world_ == 1 || !connected_) return {0, 0};
-  FlagInbox* inbox = tabs_.inbox[rank_];
+  FlagInbox* inbox = tabs_[0].inbox[rank_];
+  // pipelined mode: check both slots' error words
+  if (n_slots_ > 1) {
+    HIP_CHECK(hipMemcpyAsync(h_err_, &tabs_[1].inbox[rank_]->error,
+                             2 * sizeof(uint64_t), hipMemcpyDeviceToHost,
+                             s_err_));
+    HIP_CHECK(hipStreamSynchronize(s_err_));
+    if (h_err_[0] != 0) return {h_err_[0], h_err_[1]};
+  }
   HIP_CHECK(hipMemcpyAsync(h_err_, &inbox->error, 2 * sizeof(uint64_t),
                            hipMemcpyDeviceToHost, s_err_));
   HIP_CHECK(hipStreamSynchronize(s_err_));

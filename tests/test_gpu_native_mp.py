@@ -81,6 +81,34 @@ def test_native_relay_subset():
                       timeout=300))
 
 
+def _same_tensor_repeat(rank, world):
+    """Back-to-back calls on the SAME tensor: the per-range causality of
+    the caller fence must order copy-in(k+1) after bcast-write(k)."""
+    os.environ["ADAPCC_TIMEOUT_MS"] = "20000"
+    import torch
+
+    torch.cuda.set_device(0)
+    from adapcc_amd.runtime.engine import NativeEngine
+    from adapcc_amd.strategy.partrees import synthesize_stars
+
+    eng = NativeEngine(rank, world, device=0, cap_bytes=16 << 20)
+    eng.bootstrap()
+    eng.set_strategy(synthesize_stars(world))
+    t = torch.full((500_000,), float(rank + 1), device="cuda")
+    expect = sum(range(1, world + 1)) / world
+    for _ in range(6):
+        eng.all_reduce(t, average=True)
+    eng.synchronize()
+    # after the first average all ranks hold the mean; averaging the mean
+    # is idempotent, so any ordering bug shows up as drift
+    assert torch.allclose(t, torch.full_like(t, expect)), t[:3]
+    return True
+
+
+def test_same_tensor_repeat():
+    assert all(run_mp(_same_tensor_repeat, 2, backend="gloo", timeout=180))
+
+
 def test_native_average():
     assert all(run_mp(_world_case, 2, backend="gloo",
                       args=(123_457, [], True, "float32", 2),

@@ -1,0 +1,1 @@
+"""Relay/fault coordinator: rank-0 gRPC server + per-rank clients."""

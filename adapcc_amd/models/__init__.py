@@ -1,0 +1,1 @@
+"""Workload model zoo (GPT-2, ViT, VGG, ResNet, expert-parallel MoE)."""

@@ -1,0 +1,1 @@
+"""Native extension build + fused-op wrappers (see csrc/ and fused.py)."""

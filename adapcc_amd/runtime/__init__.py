@@ -1,0 +1,1 @@
+"""Engines (native xGMI / p2p / process-group) and the DDP hook."""

@@ -1,0 +1,1 @@
+"""Utilities: metrics counters, gradient-noise-scale estimators."""

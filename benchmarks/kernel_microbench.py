@@ -3,12 +3,13 @@
 --pmc for counter evidence): fused LayerNorm fwd/bwd and the multi-source
 reduction at the GPT-2 shapes."""
 
+import os
 import sys
 import time
 
 import torch
 
-sys.path.insert(0, __file__.rsplit("/", 2)[0])
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import adapcc_amd._core as core
 from adapcc_amd.ops.fused import FusedLayerNorm

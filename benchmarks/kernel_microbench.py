@@ -16,6 +16,8 @@ from adapcc_amd.ops.fused import FusedLayerNorm
 
 
 def main():
+    if not torch.cuda.is_available():
+        raise SystemExit("kernel_microbench needs an MI355X GPU")
     rows, cols = 65536, 768
     x = torch.randn(rows, cols, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)

@@ -65,3 +65,31 @@ def test_metrics():
     assert "phase_s" in snap
     blob = m.dump(rank=3)
     assert '"rank": 3' in blob
+
+
+def test_bench_contract_2proc_cpu():
+    """bench.py is the driver contract: verify the torchrun path emits one
+    valid JSON line with the required fields (tiny model, CPU/gloo)."""
+    import json
+    import random
+
+    port = random.randint(20000, 40000)
+    env = dict(os.environ, ADAPCC_TRANSPORT="pg")
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", str(port), os.path.join(REPO, "bench.py"),
+           "--gpus", "2", "--steps", "2", "--warmup", "1", "--model", "tiny",
+           "--batch", "2", "--seq", "64"]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                         env=env, cwd=REPO)
+    assert res.returncode == 0, res.stdout[-2000:] + res.stderr[-2000:]
+    lines = [l for l in res.stdout.splitlines() if l.startswith("{")]
+    assert len(lines) == 1, res.stdout
+    out = json.loads(lines[0])
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling", "vs_baseline",
+                "dtype", "data", "config"):
+        assert key in out, key
+    assert out["n_gpus"] == 2
+    assert out["value"] > 0
+    assert out["config"]["parallelism"] == "dp2"

@@ -139,17 +139,20 @@ PYBIND11_MODULE(_core, m) {
   m.def("compute_plan",
         [](const std::vector<std::vector<int>>& parents, int rank,
            long total_elems, int esize, long chunk_bytes,
-           const std::vector<int>& active) {
+           const std::vector<int>& active,
+           const std::vector<double>& slice_weights) {
           auto shape = adapcc::TreeShape::derive(parents);
           uint64_t mask = 0;
           for (int r : active) mask |= (1ull << r);
           if (active.empty()) mask = (1ull << shape.world) - 1;
           return plan_to_dict(adapcc::build_plan(shape, rank, total_elems,
-                                                 esize, chunk_bytes, mask));
+                                                 esize, chunk_bytes, mask,
+                                                 slice_weights));
         },
         py::arg("parents"), py::arg("rank"), py::arg("total_elems"),
         py::arg("esize"), py::arg("chunk_bytes"),
-        py::arg("active") = std::vector<int>{});
+        py::arg("active") = std::vector<int>{},
+        py::arg("slice_weights") = std::vector<double>{});
 
   m.def("compute_primitive_plan",
         [](const std::string& prim, int world, int rank, long elems, int esize,
@@ -203,7 +206,8 @@ PYBIND11_MODULE(_core, m) {
            },
            py::arg("handles"), py::arg("peer_devices") = std::vector<int>{})
       .def("set_strategy", &Engine::set_strategy, py::arg("parents"),
-           py::arg("chunk_bytes"))
+           py::arg("chunk_bytes"),
+           py::arg("slice_weights") = std::vector<double>{})
       .def("connect_local", &Engine::connect_local, py::arg("peer_addrs"))
       .def("region_addr", &Engine::region_addr)
       .def("allreduce",

@@ -30,7 +30,8 @@ class Engine {
   void connect_local(const std::vector<uintptr_t>& peer_addrs);
   uintptr_t region_addr() const { return (uintptr_t)region_; }
   void set_strategy(const std::vector<std::vector<int>>& parents,
-                    long chunk_bytes);
+                    long chunk_bytes,
+                    const std::vector<double>& slice_weights = {});
 
   // Enqueue collectives on the caller stream (asynchronous; the caller
   // stream is made to depend on completion). active_ranks empty => all.
@@ -121,6 +122,7 @@ class Engine {
   int num_trees_ = 0;
   long chunk_bytes_ = 4 * 1024 * 1024;
   TreeShape shape_;
+  std::vector<double> slice_weights_;
 
   int wgs_per_group_ = 8;
   int n_groups_ = 16;

@@ -191,12 +191,17 @@ void Engine::build_tables() {
 }
 
 void Engine::set_strategy(const std::vector<std::vector<int>>& parents,
-                          long chunk_bytes) {
+                          long chunk_bytes,
+                          const std::vector<double>& slice_weights) {
   shape_ = TreeShape::derive(parents);
   if (shape_.world != world_)
     throw std::runtime_error("parent array size != world");
+  if (!slice_weights.empty() &&
+      (int)slice_weights.size() != (int)parents.size())
+    throw std::runtime_error("slice_weights length != num trees");
   num_trees_ = (int)shape_.parents.size();
   chunk_bytes_ = chunk_bytes;
+  slice_weights_ = slice_weights;
   // strategy change invalidates cached plans
   for (auto& kv : plans_) kv.second.free_device();
   plans_.clear();
@@ -222,11 +227,12 @@ Engine::Plan& Engine::get_plan(int prim, long elems, Dtype dt, RedOp op,
   PlanData pd;
   switch (prim) {
     case 0:  // allreduce
-      pd = build_plan(shape_, rank_, elems, esize, chunk_bytes_, active_mask);
+      pd = build_plan(shape_, rank_, elems, esize, chunk_bytes_, active_mask,
+                      slice_weights_);
       break;
     case 1:  // reduce
       pd = build_reduce_plan(shape_, rank_, root, elems, esize, chunk_bytes_,
-                             active_mask);
+                             active_mask, slice_weights_);
       break;
     case 2:  // broadcast
       pd = build_broadcast_plan(world_, rank_, root, elems, esize, chunk_bytes_);

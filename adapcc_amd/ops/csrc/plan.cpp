@@ -140,6 +140,44 @@ struct TC {
   long cnt;
 };
 
+// Per-tree slice ranges over [0, total): equal by default; with weights,
+// boundaries at cumulative fractions (64-element aligned, monotone,
+// degenerate/empty slices allowed). Deterministic across ranks.
+std::vector<std::pair<long, long>> make_slices(
+    int T, long total, const std::vector<double>& weights) {
+  std::vector<std::pair<long, long>> slice(T);
+  if (weights.empty()) {
+    const long per_raw = (total + T - 1) / T;
+    const long per = (per_raw + kAlignE - 1) / kAlignE * kAlignE;
+    for (int t = 0; t < T; ++t) {
+      slice[t] = {std::min((long)t * per, total),
+                  std::min((long)(t + 1) * per, total)};
+    }
+    return slice;
+  }
+  if ((int)weights.size() != T)
+    throw std::runtime_error("slice_weights length != num trees");
+  double sum = 0.0;
+  for (double w : weights) {
+    if (!(w > 0.0)) throw std::runtime_error("slice_weights must be > 0");
+    sum += w;
+  }
+  double acc = 0.0;
+  long prev = 0;
+  for (int t = 0; t < T; ++t) {
+    acc += weights[t];
+    long end = (t == T - 1)
+                   ? total
+                   : std::min<long>(
+                         total,
+                         (long)(acc / sum * total + 0.5) / kAlignE * kAlignE);
+    if (end < prev) end = prev;
+    slice[t] = {prev, end};
+    prev = end;
+  }
+  return slice;
+}
+
 // (chunk, tree)-ordered grid over per-tree slices [beg, end).
 std::vector<TC> make_grid(const std::vector<std::pair<long, long>>& slice,
                           long chunk_elems) {
@@ -221,14 +259,16 @@ void forest_reduce_phase(const TreeShape& shape, int rank,
 }  // namespace
 
 PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
-                    int esize, long chunk_bytes, uint64_t active_mask) {
+                    int esize, long chunk_bytes, uint64_t active_mask,
+                    const std::vector<double>& slice_weights) {
   const int world = shape.world;
   const int T = (int)shape.parents.size();
   auto active = mask_to_active(active_mask, world);
 
-  const long per_raw = (total_elems + T - 1) / T;
-  const long per = (per_raw + kAlignE - 1) / kAlignE * kAlignE;
-  const long chunk_elems = pick_chunk_elems(per, esize, chunk_bytes);
+  auto slice = make_slices(T, total_elems, slice_weights);
+  long max_slice = 0;
+  for (const auto& s : slice) max_slice = std::max(max_slice, s.second - s.first);
+  const long chunk_elems = pick_chunk_elems(max_slice, esize, chunk_bytes);
 
   PlanData plan;
   plan.chunk_elems = chunk_elems;
@@ -243,11 +283,6 @@ PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
     for (int c : shape.children[t][shape.roots[t]]) publish[t].push_back(c);
   }
 
-  std::vector<std::pair<long, long>> slice(T);
-  for (int t = 0; t < T; ++t) {
-    slice[t] = {std::min((long)t * per, total_elems),
-                std::min((long)(t + 1) * per, total_elems)};
-  }
   auto grid = make_grid(slice, chunk_elems);
   forest_reduce_phase(shape, rank, grid, ana, active, publish, plan);
 
@@ -281,15 +316,17 @@ PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
 
 PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
                            long total_elems, int esize, long chunk_bytes,
-                           uint64_t active_mask) {
+                           uint64_t active_mask,
+                           const std::vector<double>& slice_weights) {
   const int world = shape.world;
   if (root < 0 || root >= world) throw std::runtime_error("reduce: bad root");
   const int T = (int)shape.parents.size();
   auto active = mask_to_active(active_mask, world);
 
-  const long per_raw = (total_elems + T - 1) / T;
-  const long per = (per_raw + kAlignE - 1) / kAlignE * kAlignE;
-  const long chunk_elems = pick_chunk_elems(per, esize, chunk_bytes);
+  auto slice = make_slices(T, total_elems, slice_weights);
+  long max_slice = 0;
+  for (const auto& s : slice) max_slice = std::max(max_slice, s.second - s.first);
+  const long chunk_elems = pick_chunk_elems(max_slice, esize, chunk_bytes);
 
   PlanData plan;
   plan.chunk_elems = chunk_elems;
@@ -300,11 +337,6 @@ PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
     ana[t].root = shape.roots[t];
     analyze_subtree(shape.roots[t], shape.children[t], active, ana[t]);
     publish[t] = {root};  // each tree's result goes to THE root only
-  }
-  std::vector<std::pair<long, long>> slice(T);
-  for (int t = 0; t < T; ++t) {
-    slice[t] = {std::min((long)t * per, total_elems),
-                std::min((long)(t + 1) * per, total_elems)};
   }
   auto grid = make_grid(slice, chunk_elems);
   forest_reduce_phase(shape, rank, grid, ana, active, publish, plan);

@@ -35,12 +35,14 @@ struct PlanData {
 // every rank derives the same global (chunk, tree) grid, so flag pushes and
 // waits agree across ranks by construction.
 PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
-                    int esize, long chunk_bytes, uint64_t active_mask);
+                    int esize, long chunk_bytes, uint64_t active_mask,
+                    const std::vector<double>& slice_weights = {});
 
 // Remaining primitives (single-node direct algorithms; see plan.cpp).
 PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
                            long total_elems, int esize, long chunk_bytes,
-                           uint64_t active_mask);
+                           uint64_t active_mask,
+                           const std::vector<double>& slice_weights = {});
 PlanData build_broadcast_plan(int world, int rank, int root, long total_elems,
                               int esize, long chunk_bytes);
 PlanData build_allgather_plan(int world, int rank, long in_elems, int esize,

@@ -109,7 +109,8 @@ class NativeEngine:
 
     def set_strategy(self, strategy: Strategy) -> None:
         parents = strategy_parent_arrays(strategy, self.world_size)
-        self._eng.set_strategy(parents, strategy.chunk_bytes)
+        weights = list(strategy.slice_weights or [])
+        self._eng.set_strategy(parents, strategy.chunk_bytes, weights)
         self._strategy_set = True
 
     # -- collectives -------------------------------------------------------

@@ -54,6 +54,7 @@ class P2PTreeEngine:
     def set_strategy(self, strategy: Strategy) -> None:
         self.parents = strategy_parent_arrays(strategy, self.world_size)
         self.chunk_bytes = strategy.chunk_bytes
+        self.slice_weights = list(strategy.slice_weights or [])
 
     # ------------------------------------------------------------------
 
@@ -69,7 +70,7 @@ class P2PTreeEngine:
         act = sorted(active) if active else list(range(self.world_size))
         plan = self.core.compute_plan(
             self.parents, self.rank, tensor.numel(), tensor.element_size(),
-            self.chunk_bytes, act)
+            self.chunk_bytes, act, self.slice_weights)
         self._execute(plan, tensor, tensor, act, average)
         return tensor
 

@@ -138,8 +138,31 @@ class MilpSolver:
                     cand.chunk_bytes = cb
                     best = cand
         assert best is not None
+        self._set_slice_weights(best)
         best = self._maybe_gurobi_refine(best)
         return best
+
+    def _tree_bottleneck_bw(self, tree) -> float:
+        """Min link bandwidth over the tree's edges (both phases use the
+        same pairs in opposite directions; take the slower direction)."""
+        edges: List[Tuple[int, int]] = []
+        self._edges(tree, edges)
+        if not edges:
+            return _DEFAULT_BW
+        return min(min(self._bw(s, d), self._bw(d, s)) for (s, d) in edges)
+
+    def _set_slice_weights(self, strategy: Strategy) -> None:
+        """Heterogeneity adaptation (reference solver.py's per-tree data
+        split s_m): slice fractions proportional to each tree's bottleneck
+        bandwidth, so all trees finish together. Equal-weight no-op on a
+        homogeneous mesh (< 5% spread)."""
+        bws = [self._tree_bottleneck_bw(t) for t in strategy.trees]
+        lo, hi = min(bws), max(bws)
+        if hi <= 0 or (hi - lo) / hi < 0.05:
+            strategy.slice_weights = None
+            return
+        total = sum(bws)
+        strategy.slice_weights = [b / total * len(bws) for b in bws]
 
     def _maybe_gurobi_refine(self, strategy: Strategy) -> Strategy:
         """Exact per-tree data-split refinement when gurobipy is available

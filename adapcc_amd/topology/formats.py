@@ -54,6 +54,9 @@ class Strategy:
 
     trees: List[TreeNode]
     chunk_bytes: int = 4 * 1024 * 1024
+    # optional per-tree slice fractions (heterogeneous links): tree t owns
+    # weight_t / sum(weights) of every tensor; None = equal slices
+    slice_weights: Optional[List[float]] = None
 
     @property
     def num_trees(self) -> int:
@@ -78,6 +81,11 @@ class Strategy:
             raise ValueError(
                 f"strategy ranks {base} do not cover world size {world_size}"
             )
+        if self.slice_weights is not None:
+            if len(self.slice_weights) != len(self.trees):
+                raise ValueError("slice_weights length != num_trees")
+            if any(w <= 0 for w in self.slice_weights):
+                raise ValueError("slice_weights must be positive")
 
     # -- roles -------------------------------------------------------------
 
@@ -146,8 +154,11 @@ def dump_strategy(strategy: Strategy, path: str) -> None:
     root = ET.Element("trees")
     if strategy.chunk_bytes:
         root.set("chunk_bytes", str(strategy.chunk_bytes))
-    for t in strategy.trees:
-        root.append(_tree_to_xml(t, "root"))
+    for i, t in enumerate(strategy.trees):
+        el = _tree_to_xml(t, "root")
+        if strategy.slice_weights is not None:
+            el.set("weight", f"{strategy.slice_weights[i]:.6f}")
+        root.append(el)
     _indent(root)
     data = ET.tostring(root, encoding="unicode", xml_declaration=False)
     with open(path, "w") as f:
@@ -165,8 +176,12 @@ def load_strategy(path_or_text: str) -> Strategy:
     if root.tag != "trees":
         raise ValueError(f"expected <trees> root, got <{root.tag}>")
     chunk = int(root.get("chunk_bytes", 4 * 1024 * 1024))
-    trees = [_tree_from_xml(el) for el in root if el.tag == "root"]
-    return Strategy(trees=trees, chunk_bytes=chunk)
+    tree_els = [el for el in root if el.tag == "root"]
+    trees = [_tree_from_xml(el) for el in tree_els]
+    weights = None
+    if any(el.get("weight") is not None for el in tree_els):
+        weights = [float(el.get("weight", 1.0)) for el in tree_els]
+    return Strategy(trees=trees, chunk_bytes=chunk, slice_weights=weights)
 
 
 # ---------------------------------------------------------------------------

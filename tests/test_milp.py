@@ -62,3 +62,25 @@ def test_milp_degraded_link_avoids_stars_bottleneck():
     assert slow > fast
     strat = solver.optimize()
     strat.validate(4)
+
+
+def test_milp_degraded_link_sets_slice_weights():
+    """A slow 0<->1 link shrinks the slices of the trees that traverse it
+    (trees rooted at 0 and 1), leaving trees 2/3 larger (reference
+    solver.py's s_m per-tree data split)."""
+    prof = uniform_profile(4)
+    prof.bandwidth[(0, 1)] = 15.0
+    prof.bandwidth[(1, 0)] = 15.0
+    g = single_node_graph(4)
+    strat = MilpSolver(g, prof).optimize()
+    strat.validate(4)
+    assert strat.slice_weights is not None
+    w = strat.slice_weights
+    assert w[0] < w[2] and w[1] < w[2], w
+    assert abs(w[2] - w[3]) < 1e-9, w
+
+
+def test_milp_homogeneous_keeps_equal_slices():
+    g = single_node_graph(4)
+    strat = MilpSolver(g, uniform_profile(4)).optimize()
+    assert strat.slice_weights is None

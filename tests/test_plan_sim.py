@@ -380,3 +380,40 @@ def test_single_element_tensor():
 
 def test_sixteen_ranks():
     sim_allreduce(stars(16), total=2048)
+
+
+def test_weighted_slices_allreduce():
+    """Heterogeneity adaptation: non-uniform per-tree slice weights still
+    produce exact results (boundaries aligned, all elements covered)."""
+    world = 4
+    parents = stars(world)
+    weights = [4.0, 1.0, 2.0, 0.5]
+    total = 5000
+    plans = [core.compute_plan(parents, r, total, 4, 256, [], weights)
+             for r in range(world)]
+    user = rand_inputs(world, total)
+    sim = Sim(world, plans, user, total)
+    sim.run()
+    expect = np.sum(user, axis=0)
+    for r in range(world):
+        np.testing.assert_allclose(sim.out[r], expect, rtol=1e-5, atol=1e-5)
+    # slice sizes actually differ (weight 4 tree ~8x weight 0.5 tree)
+    per_tree = {}
+    for u in plans[0]["bcast"]:
+        per_tree[u["tree"]] = per_tree.get(u["tree"], 0) + u["count"]
+    assert per_tree[0] > per_tree[3] * 3, per_tree
+
+
+def test_weighted_slices_extreme_and_tiny():
+    world = 3
+    weights = [100.0, 1.0, 1.0]
+    for total in (7, 64, 700):
+        plans = [core.compute_plan(stars(world), r, total, 4, 128, [], weights)
+                 for r in range(world)]
+        user = rand_inputs(world, total)
+        sim = Sim(world, plans, user, total)
+        sim.run()
+        expect = np.sum(user, axis=0)
+        for r in range(world):
+            np.testing.assert_allclose(sim.out[r], expect, rtol=1e-5,
+                                       atol=1e-5)

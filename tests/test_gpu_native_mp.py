@@ -105,6 +105,38 @@ def _same_tensor_repeat(rank, world):
     return True
 
 
+def _weighted_slices(rank, world):
+    """Heterogeneity-adapted (non-uniform) slice weights on the native
+    engine: same numerics, different per-tree work split."""
+    os.environ["ADAPCC_TIMEOUT_MS"] = "20000"
+    import torch
+    import torch.distributed as dist
+
+    torch.cuda.set_device(0)
+    from adapcc_amd.runtime.engine import NativeEngine
+    from adapcc_amd.strategy.partrees import synthesize_stars
+
+    eng = NativeEngine(rank, world, device=0, cap_bytes=16 << 20)
+    eng.bootstrap()
+    strat = synthesize_stars(world)
+    strat.slice_weights = [1.0 + 2.0 * t for t in range(world)]
+    eng.set_strategy(strat)
+    torch.manual_seed(rank)
+    t = torch.randn(777_777, device="cuda")
+    cpu = t.cpu()
+    g = [torch.zeros_like(cpu) for _ in range(world)]
+    dist.all_gather(g, cpu)
+    eng.all_reduce(t)
+    eng.synchronize()
+    torch.testing.assert_close(t.cpu(), torch.stack(g).sum(0), rtol=1e-4,
+                               atol=1e-4)
+    return True
+
+
+def test_weighted_slices_native():
+    assert all(run_mp(_weighted_slices, 4, backend="gloo", timeout=180))
+
+
 def test_same_tensor_repeat():
     assert all(run_mp(_same_tensor_repeat, 2, backend="gloo", timeout=180))
 

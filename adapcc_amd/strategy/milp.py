@@ -165,10 +165,10 @@ class MilpSolver:
         strategy.slice_weights = [b / total * len(bws) for b in bws]
 
     def _maybe_gurobi_refine(self, strategy: Strategy) -> Strategy:
-        """Exact per-tree data-split refinement when gurobipy is available
-        (reference solver.py's s_m variables): re-balance slice fractions so
-        every tree finishes simultaneously under heterogeneous link speeds.
-        No-op without gurobi — the engine currently uses equal slices."""
+        """Exact LP refinement hook when gurobipy is available: the
+        bottleneck-proportional weights from _set_slice_weights are optimal
+        for stars; an LP could refine multi-level forests with shared
+        congested links. No-op without gurobi (optional dep, as upstream)."""
         try:
             import gurobipy  # noqa: F401
         except ImportError:

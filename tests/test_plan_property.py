@@ -37,9 +37,11 @@ def test_random_forests_allreduce(seed):
     n_active = rng.randint(1, world)
     active = sorted(rng.sample(range(world), n_active))
     average = rng.random() < 0.5
+    weights = ([rng.uniform(0.1, 5.0) for _ in range(ntrees)]
+               if rng.random() < 0.4 else [])
 
     plans = [
-        core.compute_plan(parents, r, total, 4, chunk_bytes, active)
+        core.compute_plan(parents, r, total, 4, chunk_bytes, active, weights)
         for r in range(world)
     ]
     user = rand_inputs(world, total, seed=seed)

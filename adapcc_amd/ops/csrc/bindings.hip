@@ -158,7 +158,8 @@ PYBIND11_MODULE(_core, m) {
         [](const std::string& prim, int world, int rank, long elems, int esize,
            long chunk_bytes, int root,
            const std::vector<std::vector<int>>& parents,
-           const std::vector<int>& active) {
+           const std::vector<int>& active,
+           const std::vector<double>& slice_weights) {
           uint64_t mask = 0;
           for (int r : active) mask |= (1ull << r);
           if (active.empty()) mask = (1ull << world) - 1;
@@ -167,7 +168,7 @@ PYBIND11_MODULE(_core, m) {
             auto shape = parents.empty() ? adapcc::star_shape(world)
                                          : adapcc::TreeShape::derive(parents);
             pd = adapcc::build_reduce_plan(shape, rank, root, elems, esize,
-                                           chunk_bytes, mask);
+                                           chunk_bytes, mask, slice_weights);
           } else if (prim == "broadcast") {
             pd = adapcc::build_broadcast_plan(world, rank, root, elems, esize,
                                               chunk_bytes);
@@ -188,7 +189,8 @@ PYBIND11_MODULE(_core, m) {
         py::arg("prim"), py::arg("world"), py::arg("rank"), py::arg("elems"),
         py::arg("esize"), py::arg("chunk_bytes"), py::arg("root") = 0,
         py::arg("parents") = std::vector<std::vector<int>>{},
-        py::arg("active") = std::vector<int>{});
+        py::arg("active") = std::vector<int>{},
+        py::arg("slice_weights") = std::vector<double>{});
 
   py::class_<Engine>(m, "Engine")
       .def(py::init<int, int, int, size_t, double>(), py::arg("rank"),

@@ -425,20 +425,12 @@ def test_weighted_reduce_to_root():
     total = 1200
     plans = [
         core.compute_primitive_plan("reduce", world, r, total, 4, 256,
-                                    root=root, parents=stars(world))
+                                    root=root, parents=stars(world),
+                                    slice_weights=weights)
         for r in range(world)
     ]
-    # weighted variant through the direct builder
-    import adapcc_amd._core as c
-    plans_w = []
-    for r in range(world):
-        # compute_primitive_plan lacks a weights arg; exercise via
-        # compute_plan equivalence is covered elsewhere — here assert the
-        # unweighted reduce still matches after the weighted allreduce
-        # machinery changes (regression guard)
-        plans_w.append(plans[r])
     user = rand_inputs(world, total)
-    sim = Sim(world, plans_w, user, total)
+    sim = Sim(world, plans, user, total)
     sim.run()
     np.testing.assert_allclose(sim.out[root], np.sum(user, axis=0),
                                rtol=1e-5, atol=1e-5)

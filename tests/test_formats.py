@@ -113,3 +113,12 @@ def test_shipped_strategy_files():
     for f in files:
         s = load_strategy(f)
         s.validate(len(s.ranks()))
+
+
+def test_strategy_ignores_unknown_attrs(tmp_path):
+    xml = ("<trees chunk_bytes=\'1024\' future_attr=\'x\'>"
+           "<root id=\'0\' ip=\'a\' extra=\'1\'><gpu id=\'1\' ip=\'a\'/></root>"
+           "<root id=\'1\' ip=\'a\'><gpu id=\'0\' ip=\'a\'/></root></trees>")
+    s = load_strategy(xml)
+    s.validate(2)
+    assert s.chunk_bytes == 1024

@@ -1,4 +1,8 @@
-"""Multi-process test helpers (gloo on CPU, RCCL on GPU)."""
+"""Multi-process test helpers: spawn `world` fresh interpreters joined by
+a FileStore-initialized process group (gloo on CPU; on GPU boxes the
+workers still rendezvous over gloo and drive CUDA device 0 directly, which
+is how the single-GPU multi-rank engine tests work). Results are collected
+per rank; any worker exception fails the test with its traceback."""
 
 from __future__ import annotations
 

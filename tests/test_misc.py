@@ -93,3 +93,35 @@ def test_bench_contract_2proc_cpu():
     assert out["n_gpus"] == 2
     assert out["value"] > 0
     assert out["config"]["parallelism"] == "dp2"
+
+
+def test_bench_help():
+    res = subprocess.run([sys.executable, os.path.join(REPO, "bench.py"),
+                          "--help"], capture_output=True, text=True,
+                         timeout=120)
+    assert res.returncode == 0
+    for flag in ("--gpus", "--steps", "--warmup", "--precision", "--graphs"):
+        assert flag in res.stdout
+
+
+def test_wait_time_csv_shape(tmp_path):
+    """wait_time harness emits the reference CSV shape (step,wait_ms)."""
+    import random
+
+    port = random.randint(20000, 40000)
+    out = str(tmp_path / "wt.csv")
+    env = dict(os.environ, ADAPCC_TRANSPORT="pg")
+    cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+           "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+           "--master-port", str(port),
+           os.path.join(REPO, "benchmarks", "wait_time.py"),
+           "--steps", "3", "--batch", "2", "--image_size", "32",
+           "--out", out]
+    res = subprocess.run(cmd, capture_output=True, text=True, timeout=600,
+                         env=env, cwd=REPO)
+    assert res.returncode == 0, res.stdout[-1500:] + res.stderr[-1500:]
+    rows = [l.split(",") for l in open(out).read().splitlines()]
+    assert len(rows) == 3
+    for i, (step, wait) in enumerate(rows):
+        assert int(step) == i
+        assert float(wait) >= 0

@@ -47,9 +47,16 @@ def _fuzz(rank, world, seed):
         torch.manual_seed(seed * 1000 + step * 10 + rank)
 
         if op == "restrategize":
-            strat = rng.choice(["stars", "chains"])
-            eng.set_strategy(synthesize_stars(world) if strat == "stars"
-                             else synthesize_chains(world, num_trees=2))
+            kind = rng.choice(["stars", "chains", "weighted"])
+            if kind == "stars":
+                eng.set_strategy(synthesize_stars(world))
+            elif kind == "chains":
+                eng.set_strategy(synthesize_chains(world, num_trees=2))
+            else:
+                s = synthesize_stars(world)
+                s.slice_weights = [rng.uniform(0.2, 4.0)
+                                   for _ in range(world)]
+                eng.set_strategy(s)
             continue
         if op == "allreduce":
             active = (sorted(rng.sample(range(world), rng.randint(1, world)))

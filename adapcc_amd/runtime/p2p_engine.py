@@ -84,7 +84,8 @@ class P2PTreeEngine:
         plan = self.core.compute_primitive_plan(
             "reduce", self.world_size, self.rank, tensor.numel(),
             tensor.element_size(), self.chunk_bytes, root=root,
-            parents=self.parents, active=act)
+            parents=self.parents, active=act,
+            slice_weights=self.slice_weights)
         self._execute(plan, tensor, tensor, act, average)
         return tensor
 

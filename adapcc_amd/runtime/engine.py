@@ -65,7 +65,10 @@ def strategy_parent_arrays(strategy: Strategy, world_size: int) -> List[List[int
 
 
 class NativeEngine:
-    """The hipIpc/xGMI pull-engine. One instance per process (= per GPU)."""
+    """The hipIpc/xGMI pull-engine. One instance per process (= per GPU).
+    Strategies may carry per-tree slice_weights (heterogeneous links);
+    plans are cached per (primitive, size, dtype, op, active-mask, root)
+    and invalidated on set_strategy."""
 
     def __init__(
         self,

@@ -287,7 +287,13 @@ uint64_t Engine::resolve_mask(const std::vector<int>& active_ranks) const {
   if (active_ranks.empty()) {
     mask = (world_ >= 64) ? ~0ull : ((1ull << world_) - 1);
   } else {
-    for (int r : active_ranks) mask |= (1ull << r);
+    for (int r : active_ranks) {
+      if (r < 0 || r >= world_)
+        throw std::runtime_error("active rank " + std::to_string(r) +
+                                 " out of range [0, " +
+                                 std::to_string(world_) + ")");
+      mask |= (1ull << r);
+    }
   }
   if (mask == 0) throw std::runtime_error("empty active set");
   return mask;

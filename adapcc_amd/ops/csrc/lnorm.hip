@@ -19,6 +19,9 @@
 
 #include "common.h"
 
+#include <stdexcept>
+#include <string>
+
 namespace adapcc {
 
 #define DEV_INLINE __device__ __forceinline__
@@ -305,6 +308,9 @@ static void ln_fwd_dispatch(const void* x, const void* w, const void* b,
   }
   LN_COLS_LIST(LN_FWD_CASE)
 #undef LN_FWD_CASE
+  throw std::runtime_error("ln_forward: unsupported cols=" +
+                           std::to_string(cols) +
+                           " (check ln_supported first)");
 }
 
 template <typename T>
@@ -318,9 +324,13 @@ static void ln_bwd_dispatch(const void* dy, const void* x, const void* w,
     hipLaunchKernelGGL((ln_bwd_kernel<T, C>), dim3(nblocks), block, 0, s,   \
                        (const T*)dy, (const T*)x, (const T*)w, mean, rstd,  \
                        (T*)dx, ws_gamma, ws_beta, rows);                    \
+    return;                                                                 \
   }
   LN_COLS_LIST(LN_BWD_CASE)
 #undef LN_BWD_CASE
+  throw std::runtime_error("ln_backward: unsupported cols=" +
+                           std::to_string(cols) +
+                           " (check ln_supported first)");
 }
 
 void ln_forward(int dtype, const void* x, const void* w, const void* b,

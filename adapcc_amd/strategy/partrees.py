@@ -112,6 +112,13 @@ class ParTrees:
         if len(servers) <= 1:
             gpus = servers[0].gpus() if servers else []
             world = len(gpus)
+            if gpus and sorted(gpus) != list(range(world)):
+                # synthesize_stars names ranks 0..world-1; a hand-written
+                # graph with non-zero-based ranks would silently get the
+                # wrong rank names (advisor finding, round 1).
+                raise ValueError(
+                    "single-server logical graph must use global ranks "
+                    f"0..{world - 1}, got {sorted(gpus)}")
             ips = {r: servers[0].ip for r in gpus} if servers else {}
             # Fully-connected xGMI: one star per rank keeps every directed
             # link carrying exactly one flow per phase.

@@ -41,7 +41,8 @@ def main():
 
     graph = detect_node_topology(rank, local_rank, world)
     prof = profile_links(rank, world, graph,
-                         bw_elems=args.bw_mb * (1 << 20) // 4)
+                         bw_elems=args.bw_mb * (1 << 20) // 4,
+                         concurrent=False)  # unloaded per-link numbers
     if rank == 0:
         print("peer-access matrix:", local_peer_matrix())
         if prof.bandwidth:

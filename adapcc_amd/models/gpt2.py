@@ -134,7 +134,11 @@ class GPT2(nn.Module):
         logits = self.lm_head(x)
         loss = None
         if targets is not None:
-            loss = F.cross_entropy(
+            # Fused CE (csrc/ce.hip) on GPU: one read pass fwd, one
+            # read+write pass bwd; no 6.6 GB log-softmax intermediate.
+            from ..ops.fused import fused_cross_entropy
+
+            loss = fused_cross_entropy(
                 logits.view(-1, logits.size(-1)), targets.view(-1)
             )
         return logits, loss

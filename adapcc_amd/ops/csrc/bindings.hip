@@ -80,6 +80,12 @@ void ln_backward(int dtype, const void* dy, const void* x, const void* w,
                  const float* mean, const float* rstd, void* dx,
                  float* ws_gamma, float* ws_beta, long rows, long cols,
                  int nblocks, hipStream_t stream);
+void ce_forward(int dtype, const void* logits, const void* targets,
+                float* loss, float* lse, long rows, long cols,
+                long ignore_index, hipStream_t stream);
+void ce_backward(int dtype, const void* logits, const void* targets,
+                 const float* lse, const float* gscale, void* dlogits,
+                 long rows, long cols, long ignore_index, hipStream_t stream);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -131,6 +137,30 @@ PYBIND11_MODULE(_core, m) {
                               (const void*)w, (const float*)mean,
                               (const float*)rstd, (void*)dx, (float*)wsg,
                               (float*)wsb, rows, cols, nblocks,
+                              reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+
+  m.def("ce_fwd",
+        [](int dtype, uintptr_t logits, uintptr_t targets, uintptr_t loss,
+           uintptr_t lse, long rows, long cols, long ignore_index,
+           uintptr_t stream) {
+          adapcc::ce_forward(dtype, (const void*)logits,
+                             (const void*)targets, (float*)loss, (float*)lse,
+                             rows, cols, ignore_index,
+                             reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+  m.def("ce_bwd",
+        [](int dtype, uintptr_t logits, uintptr_t targets, uintptr_t lse,
+           uintptr_t gscale, uintptr_t dlogits, long rows, long cols,
+           long ignore_index, uintptr_t stream) {
+          adapcc::ce_backward(dtype, (const void*)logits,
+                              (const void*)targets, (const float*)lse,
+                              (const float*)gscale, (void*)dlogits, rows,
+                              cols, ignore_index,
                               reinterpret_cast<hipStream_t>(stream));
           hipError_t e = hipGetLastError();
           if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));

@@ -74,6 +74,59 @@ class _FusedLayerNormFn(torch.autograd.Function):
         return dx, dgamma, dbeta, None
 
 
+class _FusedCrossEntropyFn(torch.autograd.Function):
+    """Mean cross-entropy over rows without materializing log-softmax
+    (csrc/ce.hip). Saves ~2x the logits tensor of HBM traffic plus the
+    6.6 GB log-softmax intermediate on the GPT-2 flagship path."""
+
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, targets: torch.Tensor,
+                ignore_index: int) -> torch.Tensor:
+        c = _core()
+        lc = logits.contiguous()
+        rows, cols = lc.shape
+        tc = targets.contiguous().to(torch.int64)
+        loss = torch.empty(rows, dtype=torch.float32, device=lc.device)
+        lse = torch.empty_like(loss)
+        stream = torch.cuda.current_stream(lc.device).cuda_stream
+        c.ce_fwd(_DT[lc.dtype], lc.data_ptr(), tc.data_ptr(),
+                 loss.data_ptr(), lse.data_ptr(), rows, cols, ignore_index,
+                 stream)
+        n_valid = (tc != ignore_index).sum().to(torch.float32).clamp(min=1)
+        ctx.save_for_backward(lc, tc, lse, n_valid)
+        ctx.ignore_index = ignore_index
+        return loss.sum() / n_valid
+
+    @staticmethod
+    def backward(ctx, grad_out: torch.Tensor):
+        c = _core()
+        logits, targets, lse, n_valid = ctx.saved_tensors
+        rows, cols = logits.shape
+        gscale = (grad_out.to(torch.float32) / n_valid).reshape(1).contiguous()
+        dlogits = torch.empty_like(logits)
+        stream = torch.cuda.current_stream(logits.device).cuda_stream
+        c.ce_bwd(_DT[logits.dtype], logits.data_ptr(), targets.data_ptr(),
+                 lse.data_ptr(), gscale.data_ptr(), dlogits.data_ptr(),
+                 rows, cols, ctx.ignore_index, stream)
+        return dlogits, None, None
+
+
+def fused_cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
+                        ignore_index: int = -100) -> torch.Tensor:
+    """F.cross_entropy(reduction='mean') drop-in for 2D logits; runs the
+    fused HIP kernel on GPU, falls back to torch elsewhere."""
+    if (
+        logits.is_cuda
+        and logits.dim() == 2
+        and _core()
+        and logits.dtype in _DT
+        and not torch.is_autocast_enabled()
+    ):
+        return _FusedCrossEntropyFn.apply(logits, targets, ignore_index)
+    return torch.nn.functional.cross_entropy(
+        logits.float(), targets, ignore_index=ignore_index)
+
+
 class FusedLayerNorm(nn.LayerNorm):
     """Drop-in nn.LayerNorm that runs the hand-written CDNA4 kernels when
     the shape/dtype qualify (GPU, matching weight dtype, supported width);

@@ -60,8 +60,10 @@ class CausalSelfAttention(nn.Module):
         q = q.view(B, T, self.n_head, hd).transpose(1, 2)
         k = k.view(B, T, self.n_head, hd).transpose(1, 2)
         v = v.view(B, T, self.n_head, hd).transpose(1, 2)
-        y = F.scaled_dot_product_attention(
-            q, k, v, is_causal=True,
+        from ..ops.attention import flash_attention
+
+        y = flash_attention(
+            q, k, v, causal=True,
             dropout_p=self.dropout if self.training else 0.0,
         )
         y = y.transpose(1, 2).contiguous().view(B, T, C)

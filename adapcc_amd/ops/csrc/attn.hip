@@ -1,0 +1,763 @@
+// Hand-written CDNA4 flash attention (gfx950), bf16, head_dim=64, causal.
+//
+// Replaces the stock SDPA path for the GPT-2 flagship workload, where the
+// aotriton kernels ran at ~2% of the MFMA roof (rocprof round-1 evidence:
+// attention was 27.7% of the training step). Design per the MI355X HIP
+// guide's attention recipe: per-wave 32-row Q blocks, K/V tiles staged in
+// LDS (XOR-swizzled K rows for conflict-free ds_read_b128; a custom
+// interleaved V image consumed with ds_read_b64_tr_b16 hardware transpose
+// reads), softmax kept lane-local by computing S^T = K·Q^T with
+// v_mfma_f32_32x32x16_bf16 so each lane owns one Q row's scores, online
+// softmax in the exp2 domain, and P redistributed to MFMA operand layout
+// with v_cvt_pk_bf16_f32 + v_permlane32_swap_b32 (no LDS round trip).
+//
+// Layout conventions for v_mfma_f32_32x32x16_bf16 (verified on hardware by
+// the mfma_probe test):
+//   A[32x16]: lane l holds A[i = l%32][k = 8*(l/32) + e], e = 0..7
+//   B[16x32]: lane l holds B[k = 8*(l/32) + e][j = l%32]
+//   C[32x32]: lane l holds C[(r&3) + 8*(r>>2) + 4*(l>>5)][l%32], r = 0..15
+//
+// Forward computes O^T = V^T · P^T per 32-wide KV tile so the online-max
+// state and the O accumulator stay indexed by the SAME lane-local q; the
+// backward splits FA2-style into a dq kernel (q-tile outer) and a dkv
+// kernel (kv-tile outer), both recomputing P from the saved logsumexp.
+//
+// Reference parity note: the reference (JoeyYoung/adapcc) has no attention
+// kernels (models used stock torch); this op exists to meet BASELINE.json's
+// "hot ops as hand-written MFMA/LDS kernels" requirement.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+
+namespace adapcc {
+
+namespace {
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+
+constexpr int kD = 64;      // head dim
+constexpr int kQT = 128;    // q rows per workgroup (32 per wave)
+constexpr int kKV = 32;     // kv rows per tile
+constexpr float kNegBig = -3.0e38f;
+constexpr float kLog2e = 1.4426950408889634f;
+constexpr float kLn2 = 0.6931471805599453f;
+
+__device__ __forceinline__ f32x16 mfma(bf16x8 a, bf16x8 b, f32x16 c) {
+  return __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, c, 0, 0, 0);
+}
+
+// C-tile row of accumulator register r for this lane-half (l>>5).
+__device__ __forceinline__ constexpr int crow(int r, int hi) {
+  return (r & 3) + 8 * (r >> 2) + 4 * hi;
+}
+
+// Redistribute 8 C-register f32 values (rows crow(o..o+7)) into one MFMA
+// operand fragment holding elements k = 8*(l>>5) + (0..7) of the SAME
+// lane-column: cvt_pk pairs then permlane32_swap exchanges the mismatched
+// quartets between lane halves (guide T12). The s_nop covers the
+// VALU-write -> v_permlane hazard window.
+__device__ __forceinline__ bf16x8 pack_frag(const float* p) {
+  unsigned a0, a1, b0, b1;
+  asm volatile(
+      "v_cvt_pk_bf16_f32 %0, %4, %5\n\t"
+      "v_cvt_pk_bf16_f32 %1, %6, %7\n\t"
+      "v_cvt_pk_bf16_f32 %2, %8, %9\n\t"
+      "v_cvt_pk_bf16_f32 %3, %10, %11\n\t"
+      "s_nop 1\n\t"
+      "v_permlane32_swap_b32 %0, %2\n\t"
+      "v_permlane32_swap_b32 %1, %3"
+      : "=&v"(a0), "=&v"(a1), "=&v"(b0), "=&v"(b1)
+      : "v"(p[0]), "v"(p[1]), "v"(p[2]), "v"(p[3]), "v"(p[4]), "v"(p[5]),
+        "v"(p[6]), "v"(p[7]));
+  union {
+    unsigned u[4];
+    bf16x8 f;
+  } r;
+  r.u[0] = a0;
+  r.u[1] = a1;
+  r.u[2] = b0;
+  r.u[3] = b1;
+  return r.f;
+}
+
+// ---------------------------------------------------------------------------
+// LDS images.
+//
+// Row image (K / Q / dO as direct row-major fragments): [32][64] bf16 with
+// byte ^= ((row&7)<<4) XOR swizzle, read as 16-B ds_read_b128.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ int row_img_byte(int row, int col_bf16) {
+  return (row * 128 + col_bf16 * 2) ^ ((row & 7) << 4);
+}
+
+// A-or-B fragment read from a row image: lane l takes row (l&31), 8
+// consecutive bf16 at column chunk*16 + (l>>5)*8.
+__device__ __forceinline__ bf16x8 read_row_frag(const char* img, int lane,
+                                                int chunk) {
+  const int byte = row_img_byte(lane & 31, chunk * 16 + (lane >> 5) * 8);
+  return *reinterpret_cast<const bf16x8*>(img + byte);
+}
+
+// Transpose fragments come straight from the SAME swizzled row image via
+// ds_read_b64_tr_b16. Measured instruction semantics (tr2 probe, gfx950):
+// within each 16-lane group, writing lane ids as 16g+4t+m (t,m in 0..3),
+// destination lane 16g+4t+m element j receives
+//     lds[ addr_supplied_by_lane(16g+4j+t) + m elements ]
+// (addresses 8-B aligned; each source lane names one 4-element chunk).
+// So lane s supplies the address of X[8*(s>>5) + 4r + ((s>>2)&3)]
+//                                   [db*32 + 16*((s>>4)&1) + 4*(s&3)]
+// and every lane l ends up with X[kb*16 + 8*(l>>5) + 4r + j][db*32 + (l&31)]
+// in element 4r+j -- exactly the MFMA operand fragment of X^T. The 4-elem
+// chunk is 8 B inside one 16-B slot, so the row image's XOR swizzle keeps
+// it contiguous and aligned.
+__device__ __forceinline__ int tr_addr_byte(int lane, int kb, int db, int r) {
+  const int row = kb * 16 + r * 4 + 8 * (lane >> 5) + ((lane >> 2) & 3);
+  const int col = db * 32 + 16 * ((lane >> 4) & 1) + 4 * (lane & 3);
+  return row_img_byte(row, col);
+}
+
+// Two tr reads -> one 8-element fragment: lane l holds X[8*(l>>5)+e][l&31]
+// for k in [kb*16,+16), d in [db*32,+32), X^T-fragment oriented.
+__device__ __forceinline__ bf16x8 read_tr_frag(const char* img, int lane,
+                                               int kb, int db) {
+  typedef __attribute__((address_space(3))) bf16x4 lds_v4;
+  union {
+    bf16x4 h[2];
+    bf16x8 f;
+  } r;
+  r.h[0] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_v4*)(img + tr_addr_byte(lane, kb, db, 0)));
+  r.h[1] = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (lds_v4*)(img + tr_addr_byte(lane, kb, db, 1)));
+  return r.f;
+}
+
+// ---------------------------------------------------------------------------
+// Staging: 256 threads cooperatively copy a [32][64] bf16 tile (4 KB) from
+// global (row stride `rs` elements) into an LDS image. One 16-B load + one
+// 16-B ds_write per thread.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ void stage_row_img(char* img,
+                                              const __bf16* __restrict__ g,
+                                              long rs, int rows) {
+  const int c = threadIdx.x;        // 256 chunks of 16 B
+  const int row = c >> 3;
+  const int col = (c & 7) * 8;      // bf16 column
+  uint4 v = {0u, 0u, 0u, 0u};
+  if (row < rows)
+    v = *reinterpret_cast<const uint4*>(g + (long)row * rs + col);
+  *reinterpret_cast<uint4*>(img + row_img_byte(row, col)) = v;
+}
+
+// Strides for one tensor: plane = b*sb + h*sh, row stride ss (elements).
+struct TStride {
+  long sb, sh, ss;
+};
+
+struct FaParams {
+  const __bf16* q;
+  const __bf16* k;
+  const __bf16* v;
+  __bf16* o;
+  float* lse;          // [B,H,S] natural-log row logsumexp
+  TStride qs, ks, vs, os;
+  int H, S;
+  float scale;
+};
+
+// ---------------------------------------------------------------------------
+// Forward.
+// Grid: (S/128, B*H); block 256 = 4 waves, wave w owns q rows
+// qt*128 + w*32 .. +31.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void fa_fwd_kernel(FaParams p) {
+  __shared__ __attribute__((aligned(16))) char smem[4 * 4096];
+#define kimg(i) (smem + (i) * 4096)
+#define vimg(i) (smem + 8192 + (i) * 4096)
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int qt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / p.H, h = bh % p.H;
+
+  const __bf16* qg = p.q + (long)b * p.qs.sb + (long)h * p.qs.sh;
+  const __bf16* kg = p.k + (long)b * p.ks.sb + (long)h * p.ks.sh;
+  const __bf16* vg = p.v + (long)b * p.vs.sb + (long)h * p.vs.sh;
+  __bf16* og = p.o + (long)b * p.os.sb + (long)h * p.os.sh;
+  float* lseg = p.lse + (long)bh * p.S;
+
+  const int qb = qt * kQT + wave * 32;      // this wave's first q row
+  const int qrow = qb + (lane & 31);        // this lane's q row
+
+  // Q^T B-fragments, direct from global: lane l holds
+  // Q[qrow][chunk*16 + 8*hi + e].
+  bf16x8 qf[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c)
+    qf[c] = *reinterpret_cast<const bf16x8*>(
+        qg + (long)qrow * p.qs.ss + c * 16 + hi * 8);
+
+  f32x16 acc0 = {}, acc1 = {};
+  float m = kNegBig, ssum = 0.f;
+  const float c1 = p.scale * kLog2e;
+  const int nkv = (qt + 1) * (kQT / kKV);   // causal: k < (qt+1)*128
+  // prologue stage tile 0
+  stage_row_img(kimg(0), kg, p.ks.ss, 32);
+  stage_row_img(vimg(0), vg, p.vs.ss, 32);
+  __syncthreads();
+
+  int cur = 0;
+  for (int kb = 0; kb < nkv; ++kb) {
+    // issue next tile's staging loads early (simple reg staging: the
+    // loads below read global into registers; writes go after compute)
+    uint4 nk = {0, 0, 0, 0}, nv = {0, 0, 0, 0};
+    int nrow = -1, ncol = 0, nkk = -1, nd = 0;
+    if (kb + 1 < nkv) {
+      const int c = threadIdx.x;
+      nrow = c >> 3;
+      ncol = (c & 7) * 8;
+      nkk = nrow;
+      nd = ncol;
+      const long kbb = (long)(kb + 1) * kKV;
+      nk = *reinterpret_cast<const uint4*>(kg + (kbb + nrow) * p.ks.ss + ncol);
+      nv = *reinterpret_cast<const uint4*>(vg + (kbb + nkk) * p.vs.ss + nd);
+    }
+
+    const bool active = kb * kKV <= qb + 31;
+    if (active) {
+      // S^T tile: C[k][q] = sum_d K[k][d] * Q[q][d]
+      f32x16 s = {};
+#pragma unroll
+      for (int c = 0; c < 4; ++c)
+        s = mfma(read_row_frag(kimg(cur), lane, c), qf[c], s);
+
+      // causal mask + tile max (lane-local over its 16 k-rows)
+      const int kbase = kb * kKV;
+      float tmax = kNegBig;
+      float sv[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kglob = kbase + crow(r, hi);
+        sv[r] = (kglob > qrow) ? kNegBig : s[r];
+        tmax = fmaxf(tmax, sv[r]);
+      }
+      tmax = fmaxf(tmax, __shfl_xor(tmax, 32, 64));
+      const float mnew = fmaxf(m, tmax);
+      const float alpha = __builtin_amdgcn_exp2f((m - mnew) * c1);
+      m = mnew;
+      const float mc = m * c1;
+
+      // P = exp2(S*c1 - mc); accumulate lane-partial row sum
+      float pv[16];
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        pv[r] = __builtin_amdgcn_exp2f(__builtin_fmaf(sv[r], c1, -mc));
+        psum += pv[r];
+      }
+      ssum = ssum * alpha + psum;
+
+      // rescale O accumulators
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        acc0[r] *= alpha;
+        acc1[r] *= alpha;
+      }
+
+      // redistribute P to operand fragments (k blocks 0-15 / 16-31)
+      const bf16x8 p0 = pack_frag(&pv[0]);
+      const bf16x8 p1 = pack_frag(&pv[8]);
+
+      // O^T += V^T · P^T  (A = V^T via hardware transpose reads)
+      acc0 = mfma(read_tr_frag(vimg(cur), lane, 0, 0), p0, acc0);
+      acc0 = mfma(read_tr_frag(vimg(cur), lane, 1, 0), p1, acc0);
+      acc1 = mfma(read_tr_frag(vimg(cur), lane, 0, 1), p0, acc1);
+      acc1 = mfma(read_tr_frag(vimg(cur), lane, 1, 1), p1, acc1);
+    }
+
+    if (nrow >= 0) {
+      *reinterpret_cast<uint4*>(kimg(cur ^ 1) + row_img_byte(nrow, ncol)) = nk;
+      *reinterpret_cast<uint4*>(vimg(cur ^ 1) + row_img_byte(nkk, nd)) = nv;
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: combine row-sum halves, normalize, store O and lse
+  const float stot = ssum + __shfl_xor(ssum, 32, 64);
+  const float inv = 1.f / stot;
+  if ((lane >> 5) == 0)
+    lseg[qrow] = p.scale * m + __builtin_amdgcn_logf(stot) * kLn2;
+
+  __bf16* orow = og + (long)qrow * p.os.ss;
+#pragma unroll
+  for (int db = 0; db < 2; ++db) {
+    const f32x16& a = db ? acc1 : acc0;
+#pragma unroll
+    for (int mq = 0; mq < 4; ++mq) {
+      const int d0 = db * 32 + 8 * mq + 4 * hi;
+      unsigned lo, hi2;
+      asm("v_cvt_pk_bf16_f32 %0, %2, %3\n\t"
+          "v_cvt_pk_bf16_f32 %1, %4, %5"
+          : "=&v"(lo), "=&v"(hi2)
+          : "v"(a[4 * mq] * inv), "v"(a[4 * mq + 1] * inv),
+            "v"(a[4 * mq + 2] * inv), "v"(a[4 * mq + 3] * inv));
+      uint2 st = {lo, hi2};
+      *reinterpret_cast<uint2*>(orow + d0) = st;
+    }
+  }
+}
+
+#undef kimg
+#undef vimg
+
+// ---------------------------------------------------------------------------
+// Backward dQ. Grid (S/128, B*H). Per q-tile, loop kv tiles:
+//   S^T = K·Q^T           (A=K row frags, B=Q^T frags)      lane-q local
+//   P^T = exp2(S*c1 - lse*log2e), causal-masked
+//   dP^T = V·dO^T         (A=V row frags, B=dO^T frags)
+//   dS^T = P^T * (dP^T - D[q]) * scale
+//   dQ^T += K^T·dS^T      (A=K^T tr frags, B=packed dS^T)
+// ---------------------------------------------------------------------------
+struct FaBwdParams {
+  const __bf16* q;
+  const __bf16* k;
+  const __bf16* v;
+  const __bf16* dout;
+  const float* lse;      // [B,H,S]
+  const float* delta;    // [B,H,S] rowsum(dO*O)
+  __bf16* dq;
+  __bf16* dk;
+  __bf16* dv;
+  TStride qs, ks, vs, dos_, dqs, dks, dvs;
+  int H, S;
+  float scale;
+};
+
+__global__ __launch_bounds__(256) void fa_bwd_dq_kernel(FaBwdParams p) {
+  // K row image + K tr image + V row image, double-buffered
+  __shared__ __attribute__((aligned(16))) char smem[4 * 4096];
+#define kimg(i) (smem + (i) * 4096)
+#define vimg(i) (smem + 8192 + (i) * 4096)
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int qt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / p.H, h = bh % p.H;
+
+  const __bf16* qg = p.q + (long)b * p.qs.sb + (long)h * p.qs.sh;
+  const __bf16* kg = p.k + (long)b * p.ks.sb + (long)h * p.ks.sh;
+  const __bf16* vg = p.v + (long)b * p.vs.sb + (long)h * p.vs.sh;
+  const __bf16* dog = p.dout + (long)b * p.dos_.sb + (long)h * p.dos_.sh;
+  __bf16* dqg = p.dq + (long)b * p.dqs.sb + (long)h * p.dqs.sh;
+
+  const int qb = qt * kQT + wave * 32;
+  const int qrow = qb + (lane & 31);
+
+  bf16x8 qf[4], dof[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    qf[c] = *reinterpret_cast<const bf16x8*>(
+        qg + (long)qrow * p.qs.ss + c * 16 + hi * 8);
+    dof[c] = *reinterpret_cast<const bf16x8*>(
+        dog + (long)qrow * p.dos_.ss + c * 16 + hi * 8);
+  }
+  const float lse2 = p.lse[(long)bh * p.S + qrow] * kLog2e;
+  const float dvq = p.delta[(long)bh * p.S + qrow];
+  const float c1 = p.scale * kLog2e;
+
+  f32x16 dacc0 = {}, dacc1 = {};
+
+  const int nkv = (qt + 1) * (kQT / kKV);
+  stage_row_img(kimg(0), kg, p.ks.ss, 32);
+  stage_row_img(vimg(0), vg, p.vs.ss, 32);
+  __syncthreads();
+
+  int cur = 0;
+  for (int kb = 0; kb < nkv; ++kb) {
+    uint4 nk = {0, 0, 0, 0}, nv = {0, 0, 0, 0};
+    int nrow = -1, ncol = 0;
+    if (kb + 1 < nkv) {
+      const int c = threadIdx.x;
+      nrow = c >> 3;
+      ncol = (c & 7) * 8;
+      const long kbb = (long)(kb + 1) * kKV;
+      nk = *reinterpret_cast<const uint4*>(kg + (kbb + nrow) * p.ks.ss + ncol);
+      nv = *reinterpret_cast<const uint4*>(vg + (kbb + nrow) * p.vs.ss + ncol);
+    }
+
+    const bool active = kb * kKV <= qb + 31;
+    if (active) {
+      f32x16 s = {}, dp = {};
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        s = mfma(read_row_frag(kimg(cur), lane, c), qf[c], s);
+        dp = mfma(read_row_frag(vimg(cur), lane, c), dof[c], dp);
+      }
+      const int kbase = kb * kKV;
+      float ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kglob = kbase + crow(r, hi);
+        const float pe =
+            (kglob > qrow)
+                ? 0.f
+                : __builtin_amdgcn_exp2f(__builtin_fmaf(s[r], c1, -lse2));
+        ds[r] = pe * (dp[r] - dvq) * p.scale;
+      }
+      const bf16x8 d0 = pack_frag(&ds[0]);
+      const bf16x8 d1 = pack_frag(&ds[8]);
+      dacc0 = mfma(read_tr_frag(kimg(cur), lane, 0, 0), d0, dacc0);
+      dacc0 = mfma(read_tr_frag(kimg(cur), lane, 1, 0), d1, dacc0);
+      dacc1 = mfma(read_tr_frag(kimg(cur), lane, 0, 1), d0, dacc1);
+      dacc1 = mfma(read_tr_frag(kimg(cur), lane, 1, 1), d1, dacc1);
+    }
+
+    if (nrow >= 0) {
+      *reinterpret_cast<uint4*>(kimg(cur ^ 1) + row_img_byte(nrow, ncol)) = nk;
+      *reinterpret_cast<uint4*>(vimg(cur ^ 1) + row_img_byte(nrow, ncol)) = nv;
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  __bf16* drow = dqg + (long)qrow * p.dqs.ss;
+#pragma unroll
+  for (int db = 0; db < 2; ++db) {
+    const f32x16& a = db ? dacc1 : dacc0;
+#pragma unroll
+    for (int mq = 0; mq < 4; ++mq) {
+      const int d0 = db * 32 + 8 * mq + 4 * hi;
+      unsigned lo, hi2;
+      asm("v_cvt_pk_bf16_f32 %0, %2, %3\n\t"
+          "v_cvt_pk_bf16_f32 %1, %4, %5"
+          : "=&v"(lo), "=&v"(hi2)
+          : "v"(a[4 * mq]), "v"(a[4 * mq + 1]), "v"(a[4 * mq + 2]),
+            "v"(a[4 * mq + 3]));
+      uint2 st = {lo, hi2};
+      *reinterpret_cast<uint2*>(drow + d0) = st;
+    }
+  }
+}
+
+#undef kimg
+#undef vimg
+
+// ---------------------------------------------------------------------------
+// Backward dK/dV. Grid (S/128, B*H); wave owns kv rows kt*128+w*32..+31,
+// loops q tiles >= the diagonal:
+//   S = Q·K^T   as C[q][k]: A=Q row frags (LDS), B=K direct (registers)
+//   P^T[k][q] = exp2(S*c1 - lse[q]*log2e) masked            lane-k local
+//   dP[q][k]: A=dO row frags, B=V^T direct (registers)
+//   dV += P^T·dO   (A=packed P^T, B=dO^T tr frags)
+//   dS^T = P^T*(dP-D[q])*scale;  dK += dS^T·Q (A=packed dS^T, B=Q^T tr)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(256) void fa_bwd_dkv_kernel(FaBwdParams p) {
+  // q row + q tr + dO row + dO tr images (dbuf) + lse/delta tiles
+  // images use [0,16K); the epilogue reuses [0,32K) as per-wave scratch
+  __shared__ __attribute__((aligned(16))) char smem[8 * 4096 + 512];
+#define qimg(i) (smem + (i) * 4096)
+#define doimg(i) (smem + 8192 + (i) * 4096)
+#define lsetile (reinterpret_cast<float*>(smem + 32768))
+#define dtile (reinterpret_cast<float*>(smem + 32768 + 256))
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int kt = blockIdx.x;
+  const int bh = blockIdx.y;
+  const int b = bh / p.H, h = bh % p.H;
+
+  const __bf16* qg = p.q + (long)b * p.qs.sb + (long)h * p.qs.sh;
+  const __bf16* kg = p.k + (long)b * p.ks.sb + (long)h * p.ks.sh;
+  const __bf16* vg = p.v + (long)b * p.vs.sb + (long)h * p.vs.sh;
+  const __bf16* dog = p.dout + (long)b * p.dos_.sb + (long)h * p.dos_.sh;
+  __bf16* dkg = p.dk + (long)b * p.dks.sb + (long)h * p.dks.sh;
+  __bf16* dvg = p.dv + (long)b * p.dvs.sb + (long)h * p.dvs.sh;
+  const float* lseg = p.lse + (long)bh * p.S;
+  const float* deltag = p.delta + (long)bh * p.S;
+
+  const int kbb = kt * kQT + wave * 32;     // this wave's first k row
+  const int krow = kbb + (lane & 31);       // this lane's k row
+
+  // K and V^T B-fragments direct from global: lane holds
+  // K[krow][chunk*16+8*hi+e] / V[krow][...]
+  bf16x8 kf[4], vf[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    kf[c] = *reinterpret_cast<const bf16x8*>(
+        kg + (long)krow * p.ks.ss + c * 16 + hi * 8);
+    vf[c] = *reinterpret_cast<const bf16x8*>(
+        vg + (long)krow * p.vs.ss + c * 16 + hi * 8);
+  }
+
+  const float c1 = p.scale * kLog2e;
+  f32x16 dvacc0 = {}, dvacc1 = {}, dkacc0 = {}, dkacc1 = {};
+
+  const int qt0 = kt * (kQT / kKV);         // first q tile (diagonal block)
+  const int nq = p.S / kKV;
+
+  // stage q tile qt0
+  {
+    const long qb0 = (long)qt0 * kKV;
+    stage_row_img(qimg(0), qg + qb0 * p.qs.ss, p.qs.ss, 32);
+    stage_row_img(doimg(0), dog + qb0 * p.dos_.ss, p.dos_.ss, 32);
+    if (threadIdx.x < 32) {
+      lsetile[threadIdx.x] = lseg[qb0 + threadIdx.x] * kLog2e;
+      dtile[threadIdx.x] = deltag[qb0 + threadIdx.x];
+    }
+  }
+  __syncthreads();
+
+  int cur = 0;
+  for (int qi = qt0; qi < nq; ++qi) {
+    uint4 nq4 = {0, 0, 0, 0}, ndo = {0, 0, 0, 0};
+    float nlse = 0.f, nd = 0.f;
+    int nrow = -1, ncol = 0;
+    if (qi + 1 < nq) {
+      const int c = threadIdx.x;
+      nrow = c >> 3;
+      ncol = (c & 7) * 8;
+      const long qb1 = (long)(qi + 1) * kKV;
+      nq4 = *reinterpret_cast<const uint4*>(qg + (qb1 + nrow) * p.qs.ss + ncol);
+      ndo = *reinterpret_cast<const uint4*>(dog + (qb1 + nrow) * p.dos_.ss +
+                                            ncol);
+      if (threadIdx.x < 32) {
+        nlse = lseg[qb1 + threadIdx.x] * kLog2e;
+        nd = deltag[qb1 + threadIdx.x];
+      }
+    }
+
+    const bool active = qi * kKV + 31 >= kbb;
+    if (active) {
+      f32x16 s = {}, dp = {};
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        s = mfma(read_row_frag(qimg(cur), lane, c), kf[c], s);
+        dp = mfma(read_row_frag(doimg(cur), lane, c), vf[c], dp);
+      }
+      const int qbase = qi * kKV;
+      float pv[16], ds[16];
+      float* lsec = lsetile + cur * 32;
+      float* dc = dtile + cur * 32;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qglob = qbase + crow(r, hi);
+        const float pe =
+            (qglob < krow)
+                ? 0.f
+                : __builtin_amdgcn_exp2f(
+                      __builtin_fmaf(s[r], c1, -lsec[crow(r, hi)]));
+        pv[r] = pe;
+        ds[r] = pe * (dp[r] - dc[crow(r, hi)]) * p.scale;
+      }
+      const bf16x8 p0 = pack_frag(&pv[0]);
+      const bf16x8 p1 = pack_frag(&pv[8]);
+      dvacc0 = mfma(p0, read_tr_frag(doimg(cur), lane, 0, 0), dvacc0);
+      dvacc0 = mfma(p1, read_tr_frag(doimg(cur), lane, 1, 0), dvacc0);
+      dvacc1 = mfma(p0, read_tr_frag(doimg(cur), lane, 0, 1), dvacc1);
+      dvacc1 = mfma(p1, read_tr_frag(doimg(cur), lane, 1, 1), dvacc1);
+      const bf16x8 e0 = pack_frag(&ds[0]);
+      const bf16x8 e1 = pack_frag(&ds[8]);
+      dkacc0 = mfma(e0, read_tr_frag(qimg(cur), lane, 0, 0), dkacc0);
+      dkacc0 = mfma(e1, read_tr_frag(qimg(cur), lane, 1, 0), dkacc0);
+      dkacc1 = mfma(e0, read_tr_frag(qimg(cur), lane, 0, 1), dkacc1);
+      dkacc1 = mfma(e1, read_tr_frag(qimg(cur), lane, 1, 1), dkacc1);
+    }
+
+    if (nrow >= 0) {
+      *reinterpret_cast<uint4*>(qimg(cur ^ 1) + row_img_byte(nrow, ncol)) =
+          nq4;
+      *reinterpret_cast<uint4*>(doimg(cur ^ 1) + row_img_byte(nrow, ncol)) =
+          ndo;
+      if (threadIdx.x < 32) {
+        lsetile[(cur ^ 1) * 32 + threadIdx.x] = nlse;
+        dtile[(cur ^ 1) * 32 + threadIdx.x] = nd;
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // epilogue: C[k-rows][d=lane&31] per accumulator pair -> wait for the
+  // MFMA pipeline, then round-trip through this wave's staging LDS to
+  // produce coalesced row-major bf16 stores.
+  __syncthreads();
+  float* scratch = reinterpret_cast<float*>(smem) + wave * 2048;  // 8 KB
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {  // 0: dV, 1: dK
+    const f32x16& a0 = t ? dkacc0 : dvacc0;
+    const f32x16& a1 = t ? dkacc1 : dvacc1;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      scratch[crow(r, hi) * 64 + (lane & 31)] = a0[r];
+      scratch[crow(r, hi) * 64 + 32 + (lane & 31)] = a1[r];
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt(0): wave-local LDS ordering
+    __bf16* outg = (t ? dkg : dvg);
+    const TStride& os = t ? p.dks : p.dvs;
+    const int row = lane >> 1;           // 2 lanes per k row
+    const int dhalf = (lane & 1) * 32;
+#pragma unroll
+    for (int mq = 0; mq < 8; ++mq) {
+      const float* src = scratch + row * 64 + dhalf + mq * 4;
+      unsigned lo, hi2;
+      asm("v_cvt_pk_bf16_f32 %0, %2, %3\n\t"
+          "v_cvt_pk_bf16_f32 %1, %4, %5"
+          : "=&v"(lo), "=&v"(hi2)
+          : "v"(src[0]), "v"(src[1]), "v"(src[2]), "v"(src[3]));
+      uint2 st = {lo, hi2};
+      *reinterpret_cast<uint2*>(outg + (long)(kbb + row) * os.ss + dhalf +
+                                mq * 4) = st;
+    }
+    __syncthreads();  // scratch reused for dK after dV drains
+  }
+}
+
+#undef qimg
+#undef doimg
+#undef lsetile
+#undef dtile
+
+// ---------------------------------------------------------------------------
+// MFMA layout probe: C = A·B for one 32x32x16 tile with the documented
+// fragment mappings; lets a GPU test verify the lane->element maps
+// independently of the attention kernels.
+// ---------------------------------------------------------------------------
+// tr-read probe: fill a swizzled row image with X[k][d] = k*64+d (bf16),
+// run read_tr_frag for (kb, db) and dump each lane's 8 fragment elements
+// -> out[lane][e]; a test checks lane l elem e == X[kb*16+8*(l>>5)+e][db*32+(l&31)].
+__global__ void tr_probe_kernel(float* out, int kb, int db) {
+  __shared__ __attribute__((aligned(16))) char img[4096];
+  const int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 2048; i += 64) {
+    const int k = i / 64, d = i % 64;
+    *reinterpret_cast<__bf16*>(img + row_img_byte(k, d)) =
+        (__bf16)(float)(k * 64 + d);
+  }
+  __syncthreads();
+  bf16x8 f = read_tr_frag(img, lane, kb, db);
+  union {
+    bf16x8 v;
+    __bf16 e[8];
+  } u;
+  u.v = f;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) out[lane * 8 + e] = (float)u.e[e];
+}
+
+// pack probe: lane provides p[r] = lane*100 + r; dump the packed fragment.
+__global__ void pack_probe_kernel(float* out) {
+  const int lane = threadIdx.x & 63;
+  float p[8];
+#pragma unroll
+  for (int r = 0; r < 8; ++r) p[r] = (float)(lane * 100 + r);
+  bf16x8 f = pack_frag(p);
+  union {
+    bf16x8 v;
+    __bf16 e[8];
+  } u;
+  u.v = f;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) out[lane * 8 + e] = (float)u.e[e];
+}
+
+__global__ void mfma_probe_kernel(const __bf16* a, const __bf16* b, float* c) {
+  const int lane = threadIdx.x & 63;
+  union {
+    __bf16 e[8];
+    bf16x8 f;
+  } af, bf;
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    af.e[e] = a[(lane % 32) * 16 + 8 * (lane / 32) + e];   // A[i][k]
+    bf.e[e] = b[(8 * (lane / 32) + e) * 32 + (lane % 32)]; // B[k][j]
+  }
+  f32x16 acc = {};
+  acc = mfma(af.f, bf.f, acc);
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    c[crow(r, lane >> 5) * 32 + (lane % 32)] = acc[r];
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// Host launchers (bound in bindings.hip).
+// ---------------------------------------------------------------------------
+void fa_forward(const void* q, const void* k, const void* v, void* o,
+                float* lse, int B, int H, int S, const long* strides,
+                float scale, hipStream_t stream) {
+  if (S % kQT != 0) throw std::runtime_error("fa_forward: S % 128 != 0");
+  FaParams p;
+  p.q = (const __bf16*)q;
+  p.k = (const __bf16*)k;
+  p.v = (const __bf16*)v;
+  p.o = (__bf16*)o;
+  p.lse = lse;
+  p.qs = {strides[0], strides[1], strides[2]};
+  p.ks = {strides[3], strides[4], strides[5]};
+  p.vs = {strides[6], strides[7], strides[8]};
+  p.os = {strides[9], strides[10], strides[11]};
+  p.H = H;
+  p.S = S;
+  p.scale = scale;
+  dim3 grid(S / kQT, B * H);
+  hipLaunchKernelGGL(fa_fwd_kernel, grid, dim3(256), 0, stream, p);
+}
+
+void fa_backward(const void* q, const void* k, const void* v, const void* do_,
+                 const float* lse, const float* delta, void* dq, void* dk,
+                 void* dv, int B, int H, int S, const long* strides,
+                 float scale, hipStream_t stream) {
+  if (S % kQT != 0) throw std::runtime_error("fa_backward: S % 128 != 0");
+  FaBwdParams p;
+  p.q = (const __bf16*)q;
+  p.k = (const __bf16*)k;
+  p.v = (const __bf16*)v;
+  p.dout = (const __bf16*)do_;
+  p.lse = lse;
+  p.delta = delta;
+  p.dq = (__bf16*)dq;
+  p.dk = (__bf16*)dk;
+  p.dv = (__bf16*)dv;
+  p.qs = {strides[0], strides[1], strides[2]};
+  p.ks = {strides[3], strides[4], strides[5]};
+  p.vs = {strides[6], strides[7], strides[8]};
+  p.dos_ = {strides[9], strides[10], strides[11]};
+  p.dqs = {strides[12], strides[13], strides[14]};
+  p.dks = {strides[15], strides[16], strides[17]};
+  p.dvs = {strides[18], strides[19], strides[20]};
+  p.H = H;
+  p.S = S;
+  p.scale = scale;
+  dim3 grid(S / kQT, B * H);
+  hipLaunchKernelGGL(fa_bwd_dq_kernel, grid, dim3(256), 0, stream, p);
+  hipLaunchKernelGGL(fa_bwd_dkv_kernel, grid, dim3(256), 0, stream, p);
+}
+
+void mfma_probe(const void* a, const void* b, float* c, hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const __bf16*)a, (const __bf16*)b, c);
+}
+
+void tr_probe(float* out, int kb, int db, hipStream_t stream) {
+  hipLaunchKernelGGL(tr_probe_kernel, dim3(1), dim3(64), 0, stream, out, kb,
+                     db);
+}
+
+void pack_probe(float* out, hipStream_t stream) {
+  hipLaunchKernelGGL(pack_probe_kernel, dim3(1), dim3(64), 0, stream, out);
+}
+
+}  // namespace adapcc

@@ -86,6 +86,16 @@ void ce_forward(int dtype, const void* logits, const void* targets,
 void ce_backward(int dtype, const void* logits, const void* targets,
                  const float* lse, const float* gscale, void* dlogits,
                  long rows, long cols, long ignore_index, hipStream_t stream);
+void fa_forward(const void* q, const void* k, const void* v, void* o,
+                float* lse, int B, int H, int S, const long* strides,
+                float scale, hipStream_t stream);
+void fa_backward(const void* q, const void* k, const void* v, const void* do_,
+                 const float* lse, const float* delta, void* dq, void* dk,
+                 void* dv, int B, int H, int S, const long* strides,
+                 float scale, hipStream_t stream);
+void mfma_probe(const void* a, const void* b, float* c, hipStream_t stream);
+void tr_probe(float* out, int kb, int db, hipStream_t stream);
+void pack_probe(float* out, hipStream_t stream);
 }
 
 PYBIND11_MODULE(_core, m) {
@@ -162,6 +172,56 @@ PYBIND11_MODULE(_core, m) {
                               (const float*)gscale, (void*)dlogits, rows,
                               cols, ignore_index,
                               reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+
+  m.def("fa_fwd",
+        [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t o, uintptr_t lse,
+           int B, int H, int S, const std::vector<long>& strides, float scale,
+           uintptr_t stream) {
+          if (strides.size() != 12)
+            throw std::runtime_error("fa_fwd: need 12 strides");
+          adapcc::fa_forward((const void*)q, (const void*)k, (const void*)v,
+                             (void*)o, (float*)lse, B, H, S, strides.data(),
+                             scale, reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+  m.def("fa_bwd",
+        [](uintptr_t q, uintptr_t k, uintptr_t v, uintptr_t dout,
+           uintptr_t lse, uintptr_t delta, uintptr_t dq, uintptr_t dk,
+           uintptr_t dv, int B, int H, int S, const std::vector<long>& strides,
+           float scale, uintptr_t stream) {
+          if (strides.size() != 21)
+            throw std::runtime_error("fa_bwd: need 21 strides");
+          adapcc::fa_backward((const void*)q, (const void*)k, (const void*)v,
+                              (const void*)dout, (const float*)lse,
+                              (const float*)delta, (void*)dq, (void*)dk,
+                              (void*)dv, B, H, S, strides.data(), scale,
+                              reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+  m.def("mfma_probe",
+        [](uintptr_t a, uintptr_t b, uintptr_t c, uintptr_t stream) {
+          adapcc::mfma_probe((const void*)a, (const void*)b, (float*)c,
+                             reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+
+  m.def("tr_probe",
+        [](uintptr_t out, int kb, int db, uintptr_t stream) {
+          adapcc::tr_probe((float*)out, kb, db,
+                           reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+  m.def("pack_probe",
+        [](uintptr_t out, uintptr_t stream) {
+          adapcc::pack_probe((float*)out,
+                             reinterpret_cast<hipStream_t>(stream));
           hipError_t e = hipGetLastError();
           if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
         });

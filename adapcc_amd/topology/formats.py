@@ -190,11 +190,28 @@ def load_strategy(path_or_text: str) -> Strategy:
 
 
 @dataclass
+class Link:
+    """One directed GPU-GPU link as detected (xGMI on a node): peer access,
+    the detect-time bandwidth micro-probe result, and a health verdict
+    (False when the link probes well below its peers — the synthesizer
+    deweights trees that traverse unhealthy links)."""
+
+    src: int
+    dst: int
+    peer_access: bool = True
+    bw_gbps: Optional[float] = None
+    healthy: bool = True
+
+
+@dataclass
 class LogicalGraph:
-    """Cluster layout: server -> nic -> gpus (global ranks)."""
+    """Cluster layout: server -> nic -> gpus (global ranks), plus the
+    detected link structure (an MI355X extension of the reference schema;
+    files without <link> elements load with an empty link map)."""
 
     servers: List["Server"] = field(default_factory=list)
     version: str = "mi355x"
+    links: Dict[Tuple[int, int], "Link"] = field(default_factory=dict)
 
     def ranks(self) -> List[int]:
         out: List[int] = []
@@ -238,6 +255,13 @@ def dump_logical_graph(graph: LogicalGraph, path: str) -> None:
             nel = ET.SubElement(sel, "nic", {"id": str(n.nic_id)})
             for g in n.gpus:
                 ET.SubElement(nel, "gpu", {"id": str(g)})
+    for (src, dst), ln in sorted(graph.links.items()):
+        attrs = {"src": str(src), "dst": str(dst),
+                 "peer": "1" if ln.peer_access else "0",
+                 "healthy": "1" if ln.healthy else "0"}
+        if ln.bw_gbps is not None:
+            attrs["bw"] = f"{ln.bw_gbps:.2f}"
+        ET.SubElement(root, "link", attrs)
     _indent(root)
     with open(path, "w") as f:
         f.write(ET.tostring(root, encoding="unicode"))
@@ -255,6 +279,15 @@ def load_logical_graph(path: str) -> LogicalGraph:
                 nic.gpus.append(int(gel.get("id")))
             server.nics.append(nic)
         graph.servers.append(server)
+    for lel in root.findall("link"):
+        src, dst = int(lel.get("src")), int(lel.get("dst"))
+        bw = lel.get("bw")
+        graph.links[(src, dst)] = Link(
+            src=src, dst=dst,
+            peer_access=lel.get("peer", "1") == "1",
+            bw_gbps=float(bw) if bw is not None else None,
+            healthy=lel.get("healthy", "1") == "1",
+        )
     return graph
 
 

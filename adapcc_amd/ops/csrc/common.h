@@ -53,6 +53,7 @@ constexpr uint64_t kErrTimeoutReady = 1;
 constexpr uint64_t kErrTimeoutBcast = 2;
 constexpr uint64_t kErrTimeoutDone = 3;
 constexpr uint64_t kErrTimeoutCount = 4;
+constexpr uint64_t kErrStaleRead = 5;
 
 // One reduce work unit: pull nsrc buffers for (tree, chunk) and accumulate
 // into dst. Precomputed on the host per call from strategy + active set.

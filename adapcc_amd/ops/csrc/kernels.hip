@@ -374,6 +374,19 @@ __global__ void __launch_bounds__(256) bcast_kernel(
       VT::store1(dst + j, a * scale);
     }
 
+    // Staleness re-check: the end barrier only waits on the ACTIVE set,
+    // so a far-lagging excluded reader could pull from a buffer its
+    // parent already reused. The publish flag for this (tree, chunk)
+    // slot advances past args.seq exactly when that happens — detect it
+    // and raise instead of silently keeping torn data.
+    if (threadIdx.x == 0 && u.parent_rank >= 0 && u.parent_rank != me) {
+      if (peek_flag(&my_inbox->bcast[u.tree][u.chunk]) > args.seq) {
+        post_error(my_inbox, kErrStaleRead,
+                   ((uint64_t)u.tree << 32) | (uint32_t)u.chunk);
+        return;
+      }
+    }
+
     if (u.forward) {
       const bool last = unit_arrive(&counters[ui], wgs_per_group);
       if (last && threadIdx.x == 0) {
@@ -396,8 +409,9 @@ __global__ void barrier_kernel(DevTables tabs, CallArgs args, int me, int world,
   FlagInbox* my_inbox = tabs.inbox[me];
   if (blockIdx.x != 0) return;
   if (threadIdx.x == 0) {
-    for (int i = 0; i < nranks; ++i) {
-      const int r = ranks[i];
+    // signal EVERY rank (an excluded relay may be waiting to rejoin);
+    // wait only on the plan's wait set below
+    for (int r = 0; r < world; ++r) {
       if (r != me) push_flag(&tabs.inbox[r]->done[me], args.seq);
     }
   }

@@ -258,12 +258,46 @@ void forest_reduce_phase(const TreeShape& shape, int rank,
 
 }  // namespace
 
+
+// Partial active set: collapse each tree to a star rooted at an ACTIVE
+// rank, so no excluded rank aggregates or forwards (an excluded straggler
+// would otherwise sit on the actives' critical path; round-1 verdict
+// item 8). Excluded ranks stay in the tree as leaves: they contribute
+// nothing to the reduce (analyze_subtree skips them) but can still pull
+// the result in the broadcast phase. Full-mask plans keep the strategy's
+// shapes unchanged. `prefer_root`: re-root target (-1 = lowest active).
+TreeShape effective_shape(const TreeShape& shape,
+                          const std::vector<char>& active,
+                          int prefer_root = -1) {
+  const int world = shape.world;
+  bool full = true;
+  for (int r = 0; r < world; ++r)
+    if (!active[r]) { full = false; break; }
+  if (full) return shape;
+  int fallback = prefer_root;
+  if (fallback < 0 || !active[fallback]) {
+    fallback = -1;
+    for (int r = 0; r < world; ++r)
+      if (active[r]) { fallback = r; break; }
+  }
+  if (fallback < 0) return shape;  // no active rank: leave as-is
+  const int T = (int)shape.parents.size();
+  std::vector<std::vector<int>> par(T, std::vector<int>(world));
+  for (int t = 0; t < T; ++t) {
+    int root = shape.roots[t];
+    if (!active[root]) root = fallback;
+    for (int r = 0; r < world; ++r) par[t][r] = (r == root) ? -1 : root;
+  }
+  return TreeShape::derive(par);
+}
+
 PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
                     int esize, long chunk_bytes, uint64_t active_mask,
                     const std::vector<double>& slice_weights) {
   const int world = shape.world;
-  const int T = (int)shape.parents.size();
   auto active = mask_to_active(active_mask, world);
+  const TreeShape eff = effective_shape(shape, active);
+  const int T = (int)eff.parents.size();
 
   auto slice = make_slices(T, total_elems, slice_weights);
   long max_slice = 0;
@@ -276,15 +310,15 @@ PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
   std::vector<TreeAnalysis> ana(T);
   std::vector<std::vector<int>> publish(T);
   for (int t = 0; t < T; ++t) {
-    ana[t].root = shape.roots[t];
-    analyze_subtree(shape.roots[t], shape.children[t], active, ana[t]);
+    ana[t].root = eff.roots[t];
+    analyze_subtree(eff.roots[t], eff.children[t], active, ana[t]);
     // allreduce: root publishes to itself + its direct children
-    publish[t].push_back(shape.roots[t]);
-    for (int c : shape.children[t][shape.roots[t]]) publish[t].push_back(c);
+    publish[t].push_back(eff.roots[t]);
+    for (int c : eff.children[t][eff.roots[t]]) publish[t].push_back(c);
   }
 
   auto grid = make_grid(slice, chunk_elems);
-  forest_reduce_phase(shape, rank, grid, ana, active, publish, plan);
+  forest_reduce_phase(eff, rank, grid, ana, active, publish, plan);
 
   // broadcast phase: every rank receives every chunk
   for (const auto& tc : grid) {
@@ -299,11 +333,11 @@ PlanData build_plan(const TreeShape& shape, int rank, long total_elems,
       bu.parent_rank = -1;
       bu.parent_kind = (uint8_t)BufKind::Acc;
     } else {
-      int p = shape.parents[tc.t][rank];
+      int p = eff.parents[tc.t][rank];
       bu.parent_rank = p;
       bu.parent_kind = (uint8_t)(p == A.root ? BufKind::Acc : BufKind::Result);
     }
-    const auto& kids = shape.children[tc.t][rank];
+    const auto& kids = eff.children[tc.t][rank];
     if (rank != A.root && !kids.empty()) {
       bu.forward = 1;
       for (int c : kids) bu.child_rank[bu.nchildren++] = c;
@@ -320,8 +354,9 @@ PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
                            const std::vector<double>& slice_weights) {
   const int world = shape.world;
   if (root < 0 || root >= world) throw std::runtime_error("reduce: bad root");
-  const int T = (int)shape.parents.size();
   auto active = mask_to_active(active_mask, world);
+  const TreeShape eff = effective_shape(shape, active, root);
+  const int T = (int)eff.parents.size();
 
   auto slice = make_slices(T, total_elems, slice_weights);
   long max_slice = 0;
@@ -334,12 +369,12 @@ PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
   std::vector<TreeAnalysis> ana(T);
   std::vector<std::vector<int>> publish(T);
   for (int t = 0; t < T; ++t) {
-    ana[t].root = shape.roots[t];
-    analyze_subtree(shape.roots[t], shape.children[t], active, ana[t]);
+    ana[t].root = eff.roots[t];
+    analyze_subtree(eff.roots[t], eff.children[t], active, ana[t]);
     publish[t] = {root};  // each tree's result goes to THE root only
   }
   auto grid = make_grid(slice, chunk_elems);
-  forest_reduce_phase(shape, rank, grid, ana, active, publish, plan);
+  forest_reduce_phase(eff, rank, grid, ana, active, publish, plan);
 
   if (rank == root) {
     for (const auto& tc : grid) {
@@ -349,7 +384,7 @@ PlanData build_reduce_plan(const TreeShape& shape, int rank, int root,
       bu.src_offset_elems = tc.off;
       bu.dst_offset_elems = tc.off;
       bu.count_elems = tc.cnt;
-      int troot = shape.roots[tc.t];
+      int troot = eff.roots[tc.t];
       bu.parent_rank = (troot == rank) ? -1 : troot;
       bu.parent_kind = (uint8_t)BufKind::Acc;
       plan.bunits.push_back(bu);

@@ -158,3 +158,85 @@ def _relay_e2e(rank, world):
 
 def test_relay_e2e_straggler():
     assert all(run_mp(_relay_e2e, 2, backend="gloo", timeout=120))
+
+
+def test_rent_or_buy_closes_early_when_uneconomical():
+    """Round-1 verdict item 7: with a cost model (bucket size + bandwidth),
+    the first arriver must close the active set as soon as continuing to
+    wait is provably worse than excluding the stragglers — well before the
+    fixed relay threshold."""
+    from adapcc_amd.coordinator.server import CoordinatorServicer
+
+    # world 8, 100 MB buckets at 150 GB/s: full-world allreduce ~9.3 ms;
+    # threshold 10 s would otherwise keep the set open ~forever
+    svc = CoordinatorServicer(world_size=8, time_slot=0.005,
+                              relay_threshold=10.0,
+                              comm_bytes=100e6, comm_bw=150e9)
+    t0 = time.monotonic()
+    # two workers arrive; the rest never do
+    results = {}
+
+    def first():
+        results["first"] = svc.hook_fetch({"step": 0, "rank": 0})
+
+    th = threading.Thread(target=first)
+    th.start()
+    time.sleep(0.02)
+    results["second"] = svc.hook_fetch({"step": 0, "rank": 1})
+    th.join(timeout=5)
+    elapsed = time.monotonic() - t0
+    assert not th.is_alive()
+    assert elapsed < 2.0, f"set should close early, took {elapsed:.2f}s"
+    assert sorted(results["first"]["active"]) == [0, 1]
+
+
+def test_rent_or_buy_waits_when_economical():
+    """With a huge payload, excluding half the world is costlier than
+    waiting a few slots — the set must stay open until everyone arrives."""
+    from adapcc_amd.coordinator.server import CoordinatorServicer
+
+    # rent = 2*2*B/bw ~= 2.7 s; buy ~= 4.0 s -> waiting stays economical
+    # until ~1.3 s, so a 0.1 s straggler must still make the set
+    svc = CoordinatorServicer(world_size=3, time_slot=0.005,
+                              relay_threshold=5.0,
+                              comm_bytes=100e9, comm_bw=150e9)
+    out = {}
+
+    def first():
+        out["r"] = svc.hook_fetch({"step": 0, "rank": 0})
+
+    th = threading.Thread(target=first)
+    th.start()
+    time.sleep(0.02)
+    svc.hook_fetch({"step": 0, "rank": 1})
+    time.sleep(0.1)  # straggler arrives late but within economic window
+    svc.hook_fetch({"step": 0, "rank": 2})
+    th.join(timeout=5)
+    assert not th.is_alive()
+    assert sorted(out["r"]["active"]) == [0, 1, 2]
+
+
+def test_closed_step_returns_snapshot_not_reopen():
+    """A >GC-window-late straggler must get the recorded snapshot (or a
+    fault status), never a freshly negotiated singleton set."""
+    from adapcc_amd.coordinator.server import CoordinatorServicer
+
+    svc = CoordinatorServicer(world_size=2, time_slot=0.001,
+                              relay_threshold=0.01)
+    for step in range(20):
+        svc.hook_fetch({"step": step, "rank": 0})
+        svc.hook_fetch({"step": step, "rank": 1})
+    # step 1 got GC'd (keep=16); the straggler must not re-open it
+    resp = svc.hook_fetch({"step": 1, "rank": 1})
+    assert resp["status"] == 0 or resp["active"] == [0, 1]
+    assert resp["active"] != [1]
+
+
+def test_hook_fetch_updates_cost_model_ema():
+    from adapcc_amd.coordinator.server import CoordinatorServicer
+
+    svc = CoordinatorServicer(world_size=1)
+    svc.hook_fetch({"step": 0, "rank": 0, "size": 1e8, "bw": 1e11})
+    assert svc.comm_bytes == 1e8 and svc.comm_bw == 1e11
+    svc.hook_fetch({"step": 1, "rank": 0, "size": 2e8, "bw": 1e11})
+    assert 1e8 < svc.comm_bytes < 2e8

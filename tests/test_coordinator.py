@@ -202,18 +202,20 @@ def test_rent_or_buy_waits_when_economical():
                               comm_bytes=100e9, comm_bw=150e9)
     out = {}
 
-    def first():
-        out["r"] = svc.hook_fetch({"step": 0, "rank": 0})
+    def arrive(rank, delay):
+        time.sleep(delay)
+        out[rank] = svc.hook_fetch({"step": 0, "rank": rank})
 
-    th = threading.Thread(target=first)
-    th.start()
-    time.sleep(0.02)
-    svc.hook_fetch({"step": 0, "rank": 1})
-    time.sleep(0.1)  # straggler arrives late but within economic window
-    svc.hook_fetch({"step": 0, "rank": 2})
-    th.join(timeout=5)
-    assert not th.is_alive()
-    assert sorted(out["r"]["active"]) == [0, 1, 2]
+    threads = [threading.Thread(target=arrive, args=(0, 0.0)),
+               threading.Thread(target=arrive, args=(1, 0.02)),
+               # straggler arrives late but within the economic window
+               threading.Thread(target=arrive, args=(2, 0.12))]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=8)
+        assert not t.is_alive()
+    assert sorted(out[0]["active"]) == [0, 1, 2]
 
 
 def test_closed_step_returns_snapshot_not_reopen():

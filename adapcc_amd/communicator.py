@@ -54,7 +54,20 @@ class CommArgs:
     # size-adaptive transport: tensors below this many bytes take the RCCL
     # collective (latency-optimal); larger ones take the tree engine
     # (bandwidth-optimal star forest). 0 disables the hybrid.
-    small_threshold: int = 64 * 1024
+    # With the single-launch fused small-message kernel (kernels.hip
+    # small_fused_kernel) the native engine owns every bucket size: the
+    # measured 4-256 KB latency is far below the RCCL collective's on the
+    # same plans (profiles/small_lat evidence), so the RCCL bypass is off
+    # by default and kept only as an operational escape hatch
+    # (ADAPCC_SMALL_THRESHOLD=<bytes> re-enables it).
+    small_threshold: int = 0
+
+    def __post_init__(self):
+        import os as _os
+
+        env = _os.environ.get("ADAPCC_SMALL_THRESHOLD")
+        if env is not None:
+            self.small_threshold = int(env)
 
     @classmethod
     def from_namespace(cls, ns) -> "CommArgs":

@@ -124,6 +124,7 @@ class Engine {
   TreeShape shape_;
   std::vector<double> slice_weights_;
 
+  long small_fused_bytes_ = 256 * 1024;
   int wgs_per_group_ = 8;
   int n_groups_ = 16;
 

@@ -53,6 +53,12 @@ void launch_collective(Dtype dt, const void* user, void* user_mut,
                        int n_groups, hipStream_t s_red, hipStream_t s_bcast);
 void launch_barrier(const DevTables& tabs, const CallArgs& args, int me, int world,
                     const int* ranks_dev, int nranks, hipStream_t stream);
+void launch_small_fused(Dtype dt, const void* user, void* user_mut,
+                        const CopyUnit* cunits, int nc,
+                        const ReduceUnit* runits, int nr,
+                        const BcastUnit* bunits, int nb, const DevTables& tabs,
+                        const CallArgs& args, int me, int world,
+                        const int* ranks_dev, int nranks, hipStream_t stream);
 
 namespace {
 constexpr int kMaxUnits = kMaxTrees * kMaxChunkSlots;
@@ -96,6 +102,8 @@ Engine::Engine(int rank, int world, int device, size_t cap_bytes,
   HIP_CHECK(hipHostMalloc(&h_err_, 2 * sizeof(uint64_t)));
   for (int r = 0; r < kMaxRanks; ++r) peer_base_[r] = nullptr;
   peer_base_[rank_] = region_;
+  if (const char* s = getenv("ADAPCC_SMALL_FUSED_BYTES"))
+    small_fused_bytes_ = atol(s);
   if (const char* s = getenv("ADAPCC_WGS_PER_GROUP")) wgs_per_group_ = atoi(s);
   if (const char* s = getenv("ADAPCC_N_GROUPS")) n_groups_ = atoi(s);
   if (wgs_per_group_ < 1) wgs_per_group_ = 1;
@@ -326,6 +334,20 @@ void Engine::enqueue(const Plan& plan, const void* in, void* out,
     HIP_CHECK(hipStreamWaitEvent(s_red_, ev_barrier_[slot], 0));
   HIP_CHECK(hipEventRecord(ev_in_, caller));
   HIP_CHECK(hipStreamWaitEvent(s_red_, ev_in_, 0));
+
+  // Small-message fast path: one fused kernel on one stream replaces the
+  // 4-launch pipeline (+ counter memsets + cross-stream events).
+  if (args.total_elems * dtype_size(args.dtype) <= small_fused_bytes_) {
+    launch_small_fused(args.dtype, in, out, plan.d_c, (int)plan.cunits.size(),
+                       plan.d_r, (int)plan.runits.size(), plan.d_b,
+                       (int)plan.bunits.size(), tabs, args, rank_, world_,
+                       plan.d_ranks, plan.nranks, s_red_);
+    HIP_CHECK(hipGetLastError());
+    HIP_CHECK(hipEventRecord(ev_bc_, s_red_));
+    HIP_CHECK(hipEventRecord(ev_barrier_[slot], s_red_));
+    HIP_CHECK(hipStreamWaitEvent(caller, ev_bc_, 0));
+    return;
+  }
 
   const size_t cu64 = sizeof(unsigned long long);
   if (!plan.cunits.empty())

@@ -81,6 +81,14 @@ DEV_INLINE bool wait_flag_ge(const uint64_t* flag, uint64_t seq,
 DEV_INLINE bool unit_arrive(unsigned long long* counter, unsigned expect) {
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every wave drains
   __syncthreads();
+  if (expect == 1) {  // single workgroup: no counter needed (fused path)
+    if (threadIdx.x == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");  // system release
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __syncthreads();
+    return true;
+  }
   __shared__ int last_sh;
   if (threadIdx.x == 0) {
     __builtin_amdgcn_fence(__ATOMIC_RELEASE, "");  // system release
@@ -179,13 +187,12 @@ DEV_INLINE float red_combine(RedOp op, float a, float b) {
 // ---------------------------------------------------------------------------
 
 template <typename T>
-__global__ void __launch_bounds__(256) copyin_kernel(
+DEV_INLINE void copyin_body(
     const T* __restrict__ user, const CopyUnit* __restrict__ units, int n_units,
-    DevTables tabs, CallArgs args, int me, int wgs_per_group, int n_groups) {
+    const DevTables& tabs, const CallArgs& args, int me, int wgs_per_group,
+    int n_groups, int group, int wg_in_group) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
-  const int group = blockIdx.x / wgs_per_group;
-  const int wg_in_group = blockIdx.x % wgs_per_group;
   T* sendbuf = (T*)tabs.send[me];
 
   for (int ui = group; ui < n_units; ui += n_groups) {
@@ -224,6 +231,15 @@ __global__ void __launch_bounds__(256) copyin_kernel(
   }
 }
 
+template <typename T>
+__global__ void __launch_bounds__(256) copyin_kernel(
+    const T* __restrict__ user, const CopyUnit* __restrict__ units, int n_units,
+    DevTables tabs, CallArgs args, int me, int wgs_per_group, int n_groups) {
+  copyin_body<T>(user, units, n_units, tabs, args, me, wgs_per_group,
+                 n_groups, blockIdx.x / wgs_per_group,
+                 blockIdx.x % wgs_per_group);
+}
+
 // ---------------------------------------------------------------------------
 // Reduce kernel: per unit (tree, chunk) pull every effective source buffer
 // over xGMI, accumulate in fp32, write local accbuf; push readiness up the
@@ -231,14 +247,12 @@ __global__ void __launch_bounds__(256) copyin_kernel(
 // ---------------------------------------------------------------------------
 
 template <typename T>
-__global__ void __launch_bounds__(256) reduce_kernel(
-    const ReduceUnit* __restrict__ units, int n_units, DevTables tabs,
-    CallArgs args, int me, int wgs_per_group, int n_groups,
-    unsigned long long* counters /* = tabs.counters + copy units */) {
+DEV_INLINE void reduce_body(
+    const ReduceUnit* __restrict__ units, int n_units, const DevTables& tabs,
+    const CallArgs& args, int me, int wgs_per_group, int n_groups,
+    unsigned long long* counters, int group, int wg_in_group) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
-  const int group = blockIdx.x / wgs_per_group;
-  const int wg_in_group = blockIdx.x % wgs_per_group;
   FlagInbox* my_inbox = tabs.inbox[me];
   const unsigned long long deadline = realtime() + args.timeout_ticks;
 
@@ -314,6 +328,16 @@ __global__ void __launch_bounds__(256) reduce_kernel(
   }
 }
 
+template <typename T>
+__global__ void __launch_bounds__(256) reduce_kernel(
+    const ReduceUnit* __restrict__ units, int n_units, DevTables tabs,
+    CallArgs args, int me, int wgs_per_group, int n_groups,
+    unsigned long long* counters /* = tabs.counters + copy units */) {
+  reduce_body<T>(units, n_units, tabs, args, me, wgs_per_group, n_groups,
+                 counters, blockIdx.x / wgs_per_group,
+                 blockIdx.x % wgs_per_group);
+}
+
 // ---------------------------------------------------------------------------
 // Broadcast/pull kernel: wait for the tree parent's publication, pull the
 // chunk over xGMI into the user tensor (scaled); intermediates forward it
@@ -321,14 +345,12 @@ __global__ void __launch_bounds__(256) reduce_kernel(
 // ---------------------------------------------------------------------------
 
 template <typename T>
-__global__ void __launch_bounds__(256) bcast_kernel(
+DEV_INLINE void bcast_body(
     T* __restrict__ user, const BcastUnit* __restrict__ units, int n_units,
-    DevTables tabs, CallArgs args, int me, int wgs_per_group, int n_groups,
-    unsigned long long* counters) {
+    const DevTables& tabs, const CallArgs& args, int me, int wgs_per_group,
+    int n_groups, unsigned long long* counters, int group, int wg_in_group) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
-  const int group = blockIdx.x / wgs_per_group;
-  const int wg_in_group = blockIdx.x % wgs_per_group;
   FlagInbox* my_inbox = tabs.inbox[me];
   const unsigned long long deadline = realtime() + args.timeout_ticks;
   const float scale = args.scale;
@@ -398,16 +420,26 @@ __global__ void __launch_bounds__(256) bcast_kernel(
   }
 }
 
+template <typename T>
+__global__ void __launch_bounds__(256) bcast_kernel(
+    T* __restrict__ user, const BcastUnit* __restrict__ units, int n_units,
+    DevTables tabs, CallArgs args, int me, int wgs_per_group, int n_groups,
+    unsigned long long* counters) {
+  bcast_body<T>(user, units, n_units, tabs, args, me, wgs_per_group,
+                n_groups, counters, blockIdx.x / wgs_per_group,
+                blockIdx.x % wgs_per_group);
+}
+
 // ---------------------------------------------------------------------------
 // End-of-call barrier kernel: publish "done reading peers" then wait for
 // every peer's done — buffers are reusable for the next call after this.
 // Runs after both the reduce and broadcast kernels (stream-ordered).
 // ---------------------------------------------------------------------------
 
-__global__ void barrier_kernel(DevTables tabs, CallArgs args, int me, int world,
-                               const int* __restrict__ ranks, int nranks) {
+DEV_INLINE void barrier_body(const DevTables& tabs, const CallArgs& args,
+                             int me, int world, const int* __restrict__ ranks,
+                             int nranks) {
   FlagInbox* my_inbox = tabs.inbox[me];
-  if (blockIdx.x != 0) return;
   if (threadIdx.x == 0) {
     // signal EVERY rank (an excluded relay may be waiting to rejoin);
     // wait only on the plan's wait set below
@@ -431,6 +463,45 @@ __global__ void barrier_kernel(DevTables tabs, CallArgs args, int me, int world,
   }
   __syncthreads();
   if (threadIdx.x == 0) __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "");
+}
+
+__global__ void barrier_kernel(DevTables tabs, CallArgs args, int me,
+                               int world, const int* __restrict__ ranks,
+                               int nranks) {
+  if (blockIdx.x != 0) return;
+  barrier_body(tabs, args, me, world, ranks, nranks);
+}
+
+// ---------------------------------------------------------------------------
+// Small-message fused collective: ONE workgroup, ONE launch runs the whole
+// copy-in -> reduce -> broadcast -> end-barrier pipeline. For sub-256 KB
+// buckets the 4-launch path is launch-latency-bound (~1.5 us per boundary
+// + event syncs); this folds it into a single kernel. unit_arrive
+// short-circuits at wgs_per_group == 1, so no counters are touched and no
+// memset precedes the launch.
+// ---------------------------------------------------------------------------
+
+template <typename T>
+__global__ void __launch_bounds__(256) small_fused_kernel(
+    const T* __restrict__ user, T* __restrict__ user_mut,
+    const CopyUnit* __restrict__ cunits, int nc,
+    const ReduceUnit* __restrict__ runits, int nr,
+    const BcastUnit* __restrict__ bunits, int nb, DevTables tabs,
+    CallArgs args, int me, int world, const int* __restrict__ ranks,
+    int nranks) {
+  if (nc > 0) {
+    copyin_body<T>(user, cunits, nc, tabs, args, me, 1, 1, 0, 0);
+    __syncthreads();
+  }
+  if (nr > 0) {
+    reduce_body<T>(runits, nr, tabs, args, me, 1, 1, nullptr, 0, 0);
+    __syncthreads();
+  }
+  if (nb > 0) {
+    bcast_body<T>(user_mut, bunits, nb, tabs, args, me, 1, 1, nullptr, 0, 0);
+    __syncthreads();
+  }
+  barrier_body(tabs, args, me, world, ranks, nranks);
 }
 
 // ---------------------------------------------------------------------------
@@ -554,6 +625,35 @@ void launch_barrier(const DevTables& tabs, const CallArgs& args, int me, int wor
                     const int* ranks_dev, int nranks, hipStream_t stream) {
   hipLaunchKernelGGL(barrier_kernel, dim3(1), dim3(256), 0, stream, tabs, args, me,
                      world, ranks_dev, nranks);
+}
+
+void launch_small_fused(Dtype dt, const void* user, void* user_mut,
+                        const CopyUnit* cunits, int nc,
+                        const ReduceUnit* runits, int nr,
+                        const BcastUnit* bunits, int nb, const DevTables& tabs,
+                        const CallArgs& args, int me, int world,
+                        const int* ranks_dev, int nranks, hipStream_t stream) {
+  const dim3 grid(1), block(256);
+  switch (dt) {
+    case Dtype::F32:
+      hipLaunchKernelGGL((small_fused_kernel<float>), grid, block, 0, stream,
+                         (const float*)user, (float*)user_mut, cunits, nc,
+                         runits, nr, bunits, nb, tabs, args, me, world,
+                         ranks_dev, nranks);
+      break;
+    case Dtype::BF16:
+      hipLaunchKernelGGL((small_fused_kernel<__hip_bfloat16>), grid, block, 0,
+                         stream, (const __hip_bfloat16*)user,
+                         (__hip_bfloat16*)user_mut, cunits, nc, runits, nr,
+                         bunits, nb, tabs, args, me, world, ranks_dev, nranks);
+      break;
+    case Dtype::F16:
+      hipLaunchKernelGGL((small_fused_kernel<__half>), grid, block, 0, stream,
+                         (const __half*)user, (__half*)user_mut, cunits, nc,
+                         runits, nr, bunits, nb, tabs, args, me, world,
+                         ranks_dev, nranks);
+      break;
+  }
 }
 
 }  // namespace adapcc

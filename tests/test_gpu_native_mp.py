@@ -581,3 +581,49 @@ def _scoped_barrier_straggler(rank, world):
 def test_scoped_barrier_straggler():
     assert all(run_mp(_scoped_barrier_straggler, 4, backend="gloo",
                       timeout=300))
+
+
+def _small_fused(rank, world):
+    """Sub-256 KB payloads take the single-launch fused kernel; numerics
+    must match across sizes, dtypes and repeats."""
+    os.environ["ADAPCC_TIMEOUT_MS"] = "20000"
+    import torch
+    import torch.distributed as dist
+
+    torch.cuda.set_device(0)
+    from adapcc_amd.runtime.engine import NativeEngine
+    from adapcc_amd.strategy.partrees import synthesize_stars
+
+    eng = NativeEngine(rank, world, device=0, cap_bytes=8 << 20)
+    eng.bootstrap()
+    eng.set_strategy(synthesize_stars(world))
+    for n in (16, 1024, 16384, 65536, 65537):
+        for rep in range(3):
+            torch.manual_seed(rank * 97 + n + rep)
+            t = torch.randn(n, device="cuda")
+            cpu = t.cpu()
+            g = [torch.zeros_like(cpu) for _ in range(world)]
+            dist.all_gather(g, cpu)
+            eng.all_reduce(t)
+            eng.synchronize()
+            torch.testing.assert_close(t.cpu(), torch.stack(g).sum(0),
+                                       rtol=1e-4, atol=1e-4)
+    # mixed small/large back-to-back (slot/event interleaving)
+    small = torch.full((4096,), float(rank), device="cuda")
+    big = torch.full((1 << 20,), float(rank), device="cuda")
+    eng.all_reduce(small)
+    eng.all_reduce(big)
+    eng.all_reduce(small)
+    eng.synchronize()
+    expect_small = sum(range(world)) * 1.0  # reduced twice? no: allreduce
+    # small was allreduced twice: first sum(ranks), then sum over ranks of
+    # that (world * sum)
+    s1 = sum(range(world))
+    torch.testing.assert_close(
+        small, torch.full_like(small, float(s1 * world)))
+    torch.testing.assert_close(big, torch.full_like(big, float(s1)))
+    return True
+
+
+def test_small_fused_collective():
+    assert all(run_mp(_small_fused, 4, backend="gloo", timeout=300))

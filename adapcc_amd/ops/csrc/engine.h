@@ -107,7 +107,7 @@ class Engine {
   size_t region_bytes_ = 0;
   size_t inbox_off_ = 0;
   size_t slot_bytes_ = 0;   // size of one call slot (pipelined mode: 2 slots)
-  int n_slots_ = 1;         // ADAPCC_PIPELINE=1 -> 2 (cross-call overlap)
+  int n_slots_ = 2;         // cross-call overlap ON (A/B-measured); ADAPCC_PIPELINE=0 -> 1
   void* peer_base_[kMaxRanks];
 
   DevTables tabs_[2] = {};

@@ -38,7 +38,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--batch", type=int, default=64, help="per-GPU micro batch")
+    p.add_argument("--batch", type=int, default=128, help="per-GPU micro batch")
     p.add_argument("--seq", type=int, default=1024)
     p.add_argument("--model", type=str, default="small",
                    choices=["tiny", "small", "medium"])

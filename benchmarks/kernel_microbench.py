@@ -29,6 +29,17 @@ def main():
         x.grad = None
     torch.cuda.synchronize()
 
+    # fused cross-entropy at the GPT-2 logits shape
+    logits = torch.randn(8192, 50257, device="cuda", dtype=torch.bfloat16,
+                         requires_grad=True)
+    tgt = torch.randint(0, 50257, (8192,), device="cuda")
+    from adapcc_amd.ops.fused import fused_cross_entropy
+    for _ in range(10):
+        loss = fused_cross_entropy(logits, tgt)
+        loss.backward()
+        logits.grad = None
+    torch.cuda.synchronize()
+
     n = 32 << 20
     srcs = [torch.randn(n // 4, device="cuda") for _ in range(8)]
     dst = torch.empty_like(srcs[0])

@@ -85,6 +85,13 @@ Engine::Engine(int rank, int world, int device, size_t cap_bytes,
   inbox_off_ = 3 * cap_bytes_;
   slot_bytes_ = inbox_off_ + align_up(sizeof(FlagInbox));
   region_bytes_ = (size_t)n_slots_ * slot_bytes_;
+  // dmabuf IPC handles for very large regions hang hipIpcOpenMemHandle on
+  // this pool (observed at 3 GB); refuse loudly instead of wedging peers.
+  if (region_bytes_ > (2ull << 30))
+    throw std::runtime_error(
+        "engine IPC region " + std::to_string(region_bytes_) +
+        " B exceeds the 2 GiB dmabuf-IPC safety bound; lower ADAPCC_BUF_CAP"
+        " or set ADAPCC_PIPELINE=0");
   HIP_CHECK(hipMalloc(&region_, region_bytes_));
   HIP_CHECK(hipMemset(region_, 0, region_bytes_));
   HIP_CHECK(hipMalloc(&counters_,

@@ -82,7 +82,12 @@ class NativeEngine:
         if device is None:
             device = torch.cuda.current_device()
         if cap_bytes is None:
-            cap_bytes = int(os.environ.get("ADAPCC_BUF_CAP", 512 * 1024 * 1024))
+            # Per-call capacity. The IPC region is n_slots * (3*cap +
+            # inbox); dmabuf hipIpcOpenMemHandle HANGS on regions >= ~2 GB
+            # on this pool (observed: 3 GB region = 512 MB cap x 2 slots),
+            # so the default stays well under that and oversized tensors
+            # take the transparent split paths instead.
+            cap_bytes = int(os.environ.get("ADAPCC_BUF_CAP", 160 * 1024 * 1024))
         if timeout_ms is None:
             timeout_ms = float(os.environ.get("ADAPCC_TIMEOUT_MS", 30000.0))
         self.rank = rank

@@ -77,7 +77,9 @@ class MLP(nn.Module):
         self.c_proj = nn.Linear(4 * cfg.n_embd, cfg.n_embd)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.c_proj(F.gelu(self.c_fc(x), approximate="tanh"))
+        from ..ops.fused import fused_gelu
+
+        return self.c_proj(fused_gelu(self.c_fc(x)))
 
 
 class Block(nn.Module):

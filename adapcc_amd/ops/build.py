@@ -16,7 +16,7 @@ import sysconfig
 PKG_DIR = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))  # adapcc_amd/
 CSRC = os.path.join(PKG_DIR, "ops", "csrc")
 SOURCES = ["kernels.hip", "engine.hip", "plan.cpp", "lnorm.hip", "ce.hip",
-           "attn.hip", "bindings.hip"]
+           "attn.hip", "gelu.hip", "bindings.hip"]
 HEADERS = ["common.h", "engine.h", "plan.h"]
 OUT_SO = os.path.join(PKG_DIR, "_core.so")
 STAMP = os.path.join(PKG_DIR, "ops", ".build_stamp")

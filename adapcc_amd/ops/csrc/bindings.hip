@@ -94,6 +94,10 @@ void fa_backward(const void* q, const void* k, const void* v, const void* do_,
                  void* dv, int B, int H, int S, const long* strides,
                  float scale, hipStream_t stream);
 void mfma_probe(const void* a, const void* b, float* c, hipStream_t stream);
+void gelu_forward(int dtype, const void* x, void* y, long n,
+                  hipStream_t stream);
+void gelu_backward(int dtype, const void* x, const void* dy, void* dx, long n,
+                   hipStream_t stream);
 void tr_probe(float* out, int kb, int db, hipStream_t stream);
 void pack_probe(float* out, hipStream_t stream);
 }
@@ -207,6 +211,23 @@ PYBIND11_MODULE(_core, m) {
         [](uintptr_t a, uintptr_t b, uintptr_t c, uintptr_t stream) {
           adapcc::mfma_probe((const void*)a, (const void*)b, (float*)c,
                              reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+
+  m.def("gelu_fwd",
+        [](int dtype, uintptr_t x, uintptr_t y, long n, uintptr_t stream) {
+          adapcc::gelu_forward(dtype, (const void*)x, (void*)y, n,
+                               reinterpret_cast<hipStream_t>(stream));
+          hipError_t e = hipGetLastError();
+          if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
+        });
+  m.def("gelu_bwd",
+        [](int dtype, uintptr_t x, uintptr_t dy, uintptr_t dx, long n,
+           uintptr_t stream) {
+          adapcc::gelu_backward(dtype, (const void*)x, (const void*)dy,
+                                (void*)dx, n,
+                                reinterpret_cast<hipStream_t>(stream));
           hipError_t e = hipGetLastError();
           if (e != hipSuccess) throw std::runtime_error(hipGetErrorString(e));
         });

@@ -127,6 +127,43 @@ def fused_cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
         logits.float(), targets, ignore_index=ignore_index)
 
 
+class _FusedGeluFn(torch.autograd.Function):
+    """tanh-GeLU with a one-exp tanh (csrc/gelu.hip). Measured ~10%
+    faster than torch's tanh-GeLU at the flagship MLP shape (both are
+    near memory-bound; the win is the shorter VALU chain)."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor) -> torch.Tensor:
+        c = _core()
+        xc = x.contiguous()
+        y = torch.empty_like(xc)
+        stream = torch.cuda.current_stream(xc.device).cuda_stream
+        c.gelu_fwd(_DT[xc.dtype], xc.data_ptr(), y.data_ptr(), xc.numel(),
+                   stream)
+        ctx.save_for_backward(xc)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        c = _core()
+        (x,) = ctx.saved_tensors
+        dyc = dy.contiguous()
+        dx = torch.empty_like(x)
+        stream = torch.cuda.current_stream(x.device).cuda_stream
+        c.gelu_bwd(_DT[x.dtype], x.data_ptr(), dyc.data_ptr(), dx.data_ptr(),
+                   x.numel(), stream)
+        return dx
+
+
+def fused_gelu(x: torch.Tensor) -> torch.Tensor:
+    """F.gelu(x, approximate='tanh') drop-in; HIP kernel on GPU, torch
+    fallback elsewhere."""
+    if x.is_cuda and _core() and x.dtype in _DT \
+            and not torch.is_autocast_enabled():
+        return _FusedGeluFn.apply(x)
+    return torch.nn.functional.gelu(x, approximate="tanh")
+
+
 class FusedLayerNorm(nn.LayerNorm):
     """Drop-in nn.LayerNorm that runs the hand-written CDNA4 kernels when
     the shape/dtype qualify (GPU, matching weight dtype, supported width);

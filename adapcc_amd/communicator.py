@@ -107,6 +107,9 @@ class Communicator:
         self.fault_worker_list: List[int] = []
         self.effective_transport: Optional[str] = None
         self._setup_done = False
+        # recorded DDP bucket sizes (bytes) — feeds the coordinator's
+        # rent-or-buy cost model (reference accumulated_size)
+        self._bucket_bytes: List[int] = []
 
         self.synthesizer = Synthesizer(
             policy=args.policy,
@@ -250,12 +253,26 @@ class Communicator:
         from .utils.metrics import GLOBAL as metrics
 
         metrics.timer_start("relay_negotiation")
-        active = self.hooker.send_ready_request(step)
+        # feed the rent-or-buy cost model: expected step payload (recorded
+        # DDP bucket sizes once stable) and the profiled link bandwidth
+        comm_bytes = float(sum(self._bucket_bytes)) if self._bucket_bytes \
+            else 0.0
+        comm_bw = self._mean_link_bw_Bps()
+        active = self.hooker.send_ready_request(
+            step, comm_bytes=comm_bytes, comm_bw=comm_bw)
         metrics.timer_stop("relay_negotiation")
         metrics.inc("relay_negotiations")
         self.active_ranks = (
             None if len(active) >= self.world_size else active
         )
+
+    def _mean_link_bw_Bps(self) -> float:
+        """Mean profiled link bandwidth in B/s (0 when unprofiled) — the
+        rent-or-buy cost model's accumulated_bandwidth analog."""
+        if self.profile_mats is None or not self.profile_mats.bandwidth:
+            return 0.0
+        vals = list(self.profile_mats.bandwidth.values())
+        return sum(vals) / len(vals) * 1e9
 
     def _setup_native(self) -> None:
         from .runtime.engine import NativeEngine

@@ -37,9 +37,17 @@ class Hooker:
         self._channel = grpc.insecure_channel(address)
         self._hook_fetch = _stub(self._channel, "hook_fetch")
 
-    def send_ready_request(self, step: int, timeout: float = 30.0) -> List[int]:
-        resp = self._hook_fetch({"step": step, "rank": self.rank},
-                                timeout=timeout)
+    def send_ready_request(self, step: int, timeout: float = 30.0,
+                           comm_bytes: float = 0.0,
+                           comm_bw: float = 0.0) -> List[int]:
+        """Fetch the active set; optionally feed the server's rent-or-buy
+        cost model with this step's expected collective size (bytes) and
+        the measured link bandwidth (B/s)."""
+        req = {"step": step, "rank": self.rank}
+        if comm_bytes > 0 and comm_bw > 0:
+            req["size"] = comm_bytes
+            req["bw"] = comm_bw
+        resp = self._hook_fetch(req, timeout=timeout)
         return list(resp["active"])
 
     def close(self) -> None:

@@ -60,8 +60,11 @@ def adapcc_allreduce_hook(
             state.comm.notify_hook_ready(state.step)
     if state.step == 1:
         # bucket layout is stable from DDP's rebuild at iteration 1 on;
-        # record it (reference log/model_bucket_info.txt)
+        # record it (reference log/model_bucket_info.txt) and feed the
+        # coordinator's rent-or-buy cost model
         state.bucket_elems.append(tensor.numel())
+        state.comm._bucket_bytes.append(
+            tensor.numel() * tensor.element_size())
     active = state.active
     inactive_bsp = (state.bsp_mode and active is not None
                     and state.comm.rank not in active)

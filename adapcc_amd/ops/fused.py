@@ -118,6 +118,8 @@ def fused_cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
     if (
         logits.is_cuda
         and logits.dim() == 2
+        and logits.shape[0] > 0
+        and logits.shape[1] > 0
         and _core()
         and logits.dtype in _DT
         and not torch.is_autocast_enabled()

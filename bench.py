@@ -72,9 +72,12 @@ def main() -> None:
         torch.cuda.set_device(device)
 
     if world > 1:
-        dist.init_process_group(
-            backend="nccl" if use_cuda else "gloo", rank=rank, world_size=world
-        )
+        # RCCL requires one rank per GPU; when ranks share a device
+        # (1-GPU rehearsals) bootstrap over gloo instead — the data plane
+        # is the native xGMI engine either way.
+        backend = "nccl" if (use_cuda and
+                             torch.cuda.device_count() >= world) else "gloo"
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
 
     torch.manual_seed(1234 + rank)
     cfg = getattr(GPT2Config, args.model)()

@@ -253,6 +253,8 @@ class MilpSolver:
             from scipy.optimize import LinearConstraint, milp
         except ImportError:
             return None
+        if self.world < 2:
+            return None
         pool = self._tree_pool()
         n = len(pool)
         # per-tree directed-link usage for reduce (child->parent) plus
@@ -275,8 +277,10 @@ class MilpSolver:
         nv = 2 * n + 1
         cost = np.zeros(nv)
         cost[2 * n] = 1.0                         # T (us)
-        eps = 1.0 + max(depths) * 0.0
-        cost[n:2 * n] = eps                       # per-tree overhead (us)
+        # per-tree flag/launch overhead (us): small enough never to beat
+        # a real bandwidth difference, large enough to break ties toward
+        # fewer trees
+        cost[n:2 * n] = 1.0
 
         A, lb, ub = [], [], []
         # T >= sum_t s_t * B * uses / bw_e  ->  sum - T <= 0

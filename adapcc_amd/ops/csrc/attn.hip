@@ -41,9 +41,7 @@ typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
 typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
-constexpr int kD = 64;      // head dim
 constexpr int kQT = 128;    // q rows per workgroup (32 per wave)
-constexpr int kKV = 32;     // kv rows per tile
 constexpr float kNegBig = -3.0e38f;
 constexpr float kLog2e = 1.4426950408889634f;
 constexpr float kLn2 = 0.6931471805599453f;
@@ -143,17 +141,6 @@ __device__ __forceinline__ bf16x8 read_tr_frag(const char* img, int lane,
 // global (row stride `rs` elements) into an LDS image. One 16-B load + one
 // 16-B ds_write per thread.
 // ---------------------------------------------------------------------------
-__device__ __forceinline__ void stage_row_img(char* img,
-                                              const __bf16* __restrict__ g,
-                                              long rs, int rows) {
-  const int c = threadIdx.x;        // 256 chunks of 16 B
-  const int row = c >> 3;
-  const int col = (c & 7) * 8;      // bf16 column
-  uint4 v = {0u, 0u, 0u, 0u};
-  if (row < rows)
-    v = *reinterpret_cast<const uint4*>(g + (long)row * rs + col);
-  *reinterpret_cast<uint4*>(img + row_img_byte(row, col)) = v;
-}
 
 // Strides for one tensor: plane = b*sb + h*sh, row stride ss (elements).
 struct TStride {

@@ -47,7 +47,6 @@ def bench_case(rank, world, payload_elems, chunk_bytes, iters=8, warmup=3):
     eng.synchronize()
     dt = (time.perf_counter() - start) / iters
     dist.barrier()
-    eng.shutdown() if hasattr(eng, "shutdown") else None
     del eng
     return dt
 

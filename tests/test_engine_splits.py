@@ -3,8 +3,6 @@ core engine records the calls; the split boundaries and staging copies are
 checked without a GPU (the numerics themselves are GPU-tested in
 test_gpu_native_mp::test_oversized_collectives_split)."""
 
-import types
-
 import pytest
 import torch
 

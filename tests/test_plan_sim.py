@@ -434,3 +434,43 @@ def test_weighted_reduce_to_root():
     sim.run()
     np.testing.assert_allclose(sim.out[root], np.sum(user, axis=0),
                                rtol=1e-5, atol=1e-5)
+
+
+def test_partial_mask_excluded_ranks_never_aggregate_or_forward():
+    """Re-rooting invariant (round-2): with a partial active set, an
+    excluded rank's plan has no copy-in and no reduce units, and no
+    ACTIVE rank's units name an excluded rank as a source, consumer, or
+    publish target (so a wedged excluded straggler sits on nobody's
+    critical path); excluded ranks may only appear as broadcast leaves
+    pulling results."""
+    import adapcc_amd._core as core
+
+    world = 8
+    # deep chains are the adversarial shape: the old behavior forwarded
+    # broadcasts through excluded intermediates
+    chains = []
+    for t in range(3):
+        order = [(t + i) % world for i in range(world)]
+        par = [-1] * world
+        for i in range(1, world):
+            par[order[i]] = order[i - 1]
+        chains.append(par)
+    active = [0, 2, 5]
+    excluded = [r for r in range(world) if r not in active]
+    for rank in range(world):
+        plan = core.compute_plan(chains, rank, 4096, 4, 1024, active)
+        if rank in excluded:
+            assert plan["copy"] == [], rank
+            assert plan["reduce"] == [], rank
+            # broadcast pulls must come from ACTIVE parents only
+            for bu in plan["bcast"]:
+                assert bu["parent"] in active, (rank, bu)
+                assert not bu["forward"], (rank, bu)
+        else:
+            for ru in plan["reduce"]:
+                for (src, _kind) in ru["srcs"]:
+                    assert src in active, (rank, ru)
+                assert all(k in range(world) for k in ru["publish_to"])
+            for cu in plan["copy"]:
+                for n in cu["notify_to"]:
+                    assert n in active, (rank, cu)

@@ -93,3 +93,25 @@ def test_degraded_link_reweights_strategy_end_to_end():
     g2 = _fake_graph_with_links(4)
     strat2 = Synthesizer(policy="par-trees").generate_strategy(graph=g2)
     assert strat2.slice_weights is None
+
+
+def test_merge_link_facts_floors_unreachable_links():
+    """No peer access + unhealthy => bandwidth floored near zero so no
+    policy routes a trunk over the dead link."""
+    g = single_node_graph(2)
+    g.links[(0, 1)] = Link(src=0, dst=1, peer_access=False, healthy=False)
+    g.links[(1, 0)] = Link(src=1, dst=0, bw_gbps=150.0, healthy=True)
+    prof = _merge_link_facts(g, None)
+    assert prof.bandwidth[(0, 1)] == pytest.approx(0.01)
+    assert prof.bandwidth[(1, 0)] == pytest.approx(150.0)
+
+
+def test_merge_keeps_minimum_of_probe_and_profile():
+    from adapcc_amd.topology.formats import ProfileMatrices
+
+    g = single_node_graph(2)
+    g.links[(0, 1)] = Link(src=0, dst=1, bw_gbps=100.0, healthy=True)
+    prof0 = ProfileMatrices()
+    prof0.bandwidth[(0, 1)] = 80.0   # profile saw it slower
+    merged = _merge_link_facts(g, prof0)
+    assert merged.bandwidth[(0, 1)] == pytest.approx(80.0)

@@ -125,3 +125,43 @@ def test_wait_time_csv_shape(tmp_path):
     for i, (step, wait) in enumerate(rows):
         assert int(step) == i
         assert float(wait) >= 0
+
+
+def test_commargs_env_override(monkeypatch):
+    from adapcc_amd.communicator import CommArgs
+
+    monkeypatch.setenv("ADAPCC_SMALL_THRESHOLD", "65536")
+    assert CommArgs().small_threshold == 65536
+    monkeypatch.delenv("ADAPCC_SMALL_THRESHOLD")
+    assert CommArgs().small_threshold == 0
+
+
+def test_fused_ops_cpu_fallbacks():
+    """On CPU every fused op must silently use the torch reference."""
+    import torch
+
+    from adapcc_amd.ops.fused import fused_cross_entropy, fused_gelu
+
+    x = torch.randn(8, 16, requires_grad=True)
+    y = fused_gelu(x)
+    torch.testing.assert_close(
+        y, torch.nn.functional.gelu(x, approximate="tanh"))
+    logits = torch.randn(4, 11, requires_grad=True)
+    tgt = torch.randint(0, 11, (4,))
+    loss = fused_cross_entropy(logits, tgt)
+    torch.testing.assert_close(
+        loss, torch.nn.functional.cross_entropy(logits, tgt))
+
+
+def test_flash_attention_cpu_fallback():
+    import torch
+
+    from adapcc_amd.ops.attention import fa_supported, flash_attention
+
+    q = torch.randn(1, 2, 128, 64)
+    k, v = torch.randn_like(q), torch.randn_like(q)
+    assert not fa_supported(q, k, v, True, 0.0)  # CPU -> no
+    out = flash_attention(q, k, v, causal=True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q, k, v, is_causal=True)
+    torch.testing.assert_close(out, ref)
